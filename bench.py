@@ -1,0 +1,234 @@
+#!/usr/bin/env python3
+"""bench.py — transpose! effective GiB/s on MI355X (BASELINE.json metric).
+
+A "step" is ONE x->y pencil transpose of the global grid (the hot path,
+Transpositions.transpose!).  Default workload: 1024^3 Float64 x->y — the
+configuration the BASELINE metric is quoted on — on N GPUs of one node with
+the BASELINE grids (N=1: 1x1, 2: 2x1, 4: 2x2, 8: 2x4).
+
+  python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run with one rank
+per GPU over RCCL; ranks read RANK/LOCAL_RANK/WORLD_SIZE from the env.
+
+value = prod(size_global) * elem_size / t_step / 2^30 (whole-job effective
+GiB/s; inputs resident in HBM when the timed region starts).  Also emitted:
+`roofline` (dominant kernel, HIP-event timed, algorithmic bytes vs the 8 TB/s
+HBM peak) and `cpu_baseline` (the C oracle restatement timed on this box's
+host cores, rank 0 at N=1 only).
+"""
+
+import argparse
+import json
+import math
+import os
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+import torch  # noqa: E402
+
+from pencilarrays_amd import Pencil, PencilArray, Topology, Transposition  # noqa: E402
+
+DEFAULT_GRIDS = {1: (1, 1), 2: (2, 1), 4: (2, 2), 8: (2, 4)}
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--size", type=int, nargs=3, default=[1024, 1024, 1024])
+    ap.add_argument("--grid", type=int, nargs=2, default=None,
+                    help="process grid P1 P2 (default: BASELINE grids)")
+    ap.add_argument("--permuted", action="store_true",
+                    help="config-5 variant: output pencil memory-permuted "
+                         "(1,2,0) — the PencilFFTs layout")
+    ap.add_argument("--dtype", default="float64",
+                    choices=["float64", "complex64"])
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    return ap.parse_args()
+
+
+def cpu_baseline(sample=(512, 512, 512), reps=3):
+    """Time the line-faithful C oracle (the reference algorithm, OpenMP over
+    this box's host cores) on a bounded sample of the same workload: world=1
+    x->y, Float64 — kind 'port' (restatement), reported as context."""
+    exe = os.path.join(REPO, "oracle", "oracle_bench")
+    if not os.path.exists(exe):
+        try:
+            subprocess.run(["make", "-C", os.path.join(REPO, "oracle"),
+                            "oracle_bench"], check=True, capture_output=True)
+        except Exception:
+            return None
+    try:
+        out = subprocess.run(
+            [exe, str(sample[0]), str(sample[1]), str(sample[2]), "1", "1",
+             str(reps)],
+            capture_output=True, text=True, timeout=600, check=True)
+        r = json.loads(out.stdout.strip().splitlines()[-1])
+        return {
+            "value": round(r["gib_per_s"], 3),
+            "unit": "GiB/s",
+            "cores": r["threads"],
+            "kind": "port",
+            "sample": f"{sample[0]}^3 Float64 x->y world=1, best of {reps} "
+                      f"(the reference algorithm incl. its staged self-copy)",
+        }
+    except Exception:
+        return None
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, args.gpus)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl", rank=rank, world_size=world)
+
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    grid = tuple(args.grid) if args.grid else DEFAULT_GRIDS.get(
+        n_gpus, (2, n_gpus // 2))
+    assert math.prod(grid) == n_gpus, f"grid {grid} != {n_gpus} ranks"
+
+    dims = tuple(args.size)
+    tdt = {"float64": torch.float64, "complex64": torch.complex64}[args.dtype]
+    esz = torch.tensor([], dtype=tdt).element_size()
+
+    topo = Topology(grid)
+    Pi = Pencil(topo, dims, (1, 2))
+    po_perm = (1, 2, 0) if args.permuted else None
+    Po = Pencil(topo, dims, (0, 2), permute=po_perm)
+
+    gen = torch.Generator(device="cpu").manual_seed(0xC0FFEE + rank)
+    n_in = Pi.length_local(rank)
+    if tdt.is_complex:
+        src_t = torch.randn(n_in, generator=gen, dtype=tdt).to(device)
+    else:
+        src_t = torch.randn(n_in, generator=gen, dtype=tdt).to(device)
+    src = PencilArray(Pi, rank, src_t)
+    dst = PencilArray(
+        Po, rank, torch.empty(Po.length_local(rank), dtype=tdt, device=device))
+
+    t = Transposition(dst, src)
+    t.execute()  # builds native plan, allocates staging, inits RCCL comms
+    torch.cuda.synchronize()
+
+    stream = torch.cuda.current_stream()
+
+    def step(sync=False):
+        t._native.execute(src.data, dst.data, sync=sync)
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+
+    # HIP events around the timed region on the launch stream: per-step
+    # kernel-side duration for the roofline (single-kernel steps at N=1).
+    ev_a = torch.cuda.Event(enable_timing=True)
+    ev_b = torch.cuda.Event(enable_timing=True)
+    ev_a.record(stream)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    ev_b.record(stream)
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+    elapsed = t1 - t0
+    gpu_ms = ev_a.elapsed_time(ev_b)
+
+    if dist:
+        tmax = torch.tensor([elapsed], device=device)
+        dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
+        elapsed = float(tmax.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    global_bytes = math.prod(dims) * esz
+    gib_s = global_bytes / (elapsed / args.steps) / 2**30
+
+    if rank != 0:
+        return
+
+    # Roofline of the dominant kernel.  At N=1 the whole step is the fused
+    # local copy: algorithmic HBM traffic = 2 * elem_size per global element
+    # (read + write; the reference's staged path moves 4x).  At
+    # N>1 the dominant resource is the xGMI link: bytes crossing one link
+    # per step = remote fraction of the local block, both directions.
+    P_sub = grid[0]
+    if n_gpus == 1:
+        kernel_ms = gpu_ms / args.steps
+        algo_bytes = 2 * global_bytes  # per launch == per step at N=1
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(algo_bytes / (kernel_ms * 1e-3) / 1e9, 1),
+            "peak": 8000.0,
+            "unit": "GB/s",
+            "frac": round(algo_bytes / (kernel_ms * 1e-3) / 8e12, 4),
+            "traffic": None,
+            "kernel": "fused local permuted copy (k_copy_1d / k_transpose_tile)",
+        }
+    else:
+        local_bytes = global_bytes // n_gpus
+        remote_frac = (P_sub - 1) / P_sub
+        link_bytes = local_bytes * remote_frac / max(P_sub - 1, 1)
+        roofline = {
+            "bound": "xgmi",
+            "achieved": round(link_bytes / (elapsed / args.steps) / 1e9, 1),
+            "peak": 153.0,
+            "unit": "GB/s",
+            "frac": round(link_bytes / (elapsed / args.steps) / 153e9, 4),
+            "traffic": None,
+            "kernel": "per-link xGMI send (grouped ncclSend/Recv)",
+        }
+
+    cb = None
+    if n_gpus == 1 and not args.no_cpu_baseline:
+        cb = cpu_baseline()
+
+    sz = "x".join(map(str, dims))
+    result = {
+        "metric": "transpose! effective GiB/s (1024^3 Float64 x->y pencil)",
+        "value": round(gib_s, 2),
+        "unit": "GiB/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 4),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "f64" if args.dtype == "float64" else "c64",
+        "data": "synthetic",
+        "config": {
+            "workload": f"{sz} {args.dtype} x->y pencil transpose"
+                        + (" (memory-permuted output, PencilFFTs layout)"
+                           if args.permuted else ""),
+            "grid": f"{grid[0]}x{grid[1]}",
+            "exchange": "rccl" if n_gpus > 1 else "none (local path)",
+        },
+        "roofline": roofline,
+        "cpu_baseline": cb,
+    }
+    print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

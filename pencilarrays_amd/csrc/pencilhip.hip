@@ -1,0 +1,1074 @@
+/* pencilhip.hip — MI355X-native (gfx950) global-transpose engine.
+ *
+ * Implements include/pencilhip.h: the engine under PencilArrays.jl's
+ * Transpositions.transpose! hot path, built from scratch for CDNA4:
+ *
+ *  - every data movement (pack, unpack, fused local/self copy) is a strided
+ *    copy descriptor executed by one of three HIP kernels chosen at plan
+ *    time: a vector 1-D copy, a batched linear-runs copy, or an LDS-tiled
+ *    N-d transpose (coalesced reads AND writes, 64-wide wavefronts);
+ *  - the subgroup exchange is grouped ncclSend/ncclRecv (RCCL over xGMI) on
+ *    the caller's HIP stream — replacing MPI.Isend/Irecv / MPI.Alltoallv!
+ *    (Transpositions.jl:419-428, 463-479);
+ *  - the self block bypasses the staging buffers entirely (the reference
+ *    stages it through recv_buf, :394-404 + :588-606 = 32 B/elem of HBM
+ *    traffic; the fused kernel moves 16 B/elem), bit-identical results.
+ *
+ * Algorithm restated from (0-based, half-open ranges):
+ *   split formula           data_ranges.jl:4-9
+ *   axes/regions            data_ranges.jl:15-45
+ *   to_local                Pencils.jl:579-587
+ *   topology / subgroups    MPITopologies.jl:125-136, 208-251
+ *   plan + peer blocks      Transpositions.jl:94-119, 282-344, 346-431,
+ *                           489-536, 542-552
+ *   pack/unpack semantics   Transpositions.jl:554-667 (copy_range! /
+ *                           copy_permuted!, perm = permutation(Po)/
+ *                           permutation(Pi) at :506)
+ *
+ * Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 -shared -fPIC
+ *        pencilhip.hip -I../../include -L/opt/rocm/lib -lrccl
+ *        -o ../libpencilhip.so
+ */
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "pencilhip.h"
+
+/* ================= error handling ================================== */
+
+static thread_local std::string g_err;
+
+extern "C" const char *pa_last_error(void) { return g_err.c_str(); }
+
+static pa_status fail(const char *fmt, ...)
+{
+    char buf[512];
+    va_list ap;
+    va_start(ap, fmt);
+    vsnprintf(buf, sizeof buf, fmt, ap);
+    va_end(ap);
+    g_err = buf;
+    return 1;
+}
+
+#define HIP_CHECK(x)                                                         \
+    do {                                                                     \
+        hipError_t err_ = (x);                                               \
+        if (err_ != hipSuccess)                                              \
+            return fail("HIP error %s at %s:%d: %s", hipGetErrorName(err_),  \
+                        __FILE__, __LINE__, hipGetErrorString(err_));        \
+    } while (0)
+
+#define NCCL_CHECK(x)                                                        \
+    do {                                                                     \
+        ncclResult_t err_ = (x);                                             \
+        if (err_ != ncclSuccess)                                             \
+            return fail("RCCL error at %s:%d: %s", __FILE__, __LINE__,       \
+                        ncclGetErrorString(err_));                           \
+    } while (0)
+
+/* ================= kernels ========================================= */
+
+#define MAXND 8
+
+struct DescDev {
+    int nd;
+    int64_t dims[MAXND], sstr[MAXND], dstr[MAXND];
+    int64_t soff, doff, total;
+};
+
+/* 1-D contiguous copy, T = 4/8/16-byte word.  The N=1 identity-permutation
+ * transpose collapses to this (a straight device copy). */
+template <typename T>
+__global__ __launch_bounds__(256) void k_copy_1d(const T *__restrict__ src,
+                                                 T *__restrict__ dst,
+                                                 int64_t n)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = src[i];
+}
+
+/* Batched linear runs: axis 0 contiguous on both sides (coalesced); outer
+ * axes strided.  Used for pack (strided window -> contiguous buffer when the
+ * fastest axis survives) and its inverse. */
+template <typename T>
+__global__ __launch_bounds__(256) void k_copy_linear(const T *__restrict__ src,
+                                                     T *__restrict__ dst,
+                                                     DescDev d)
+{
+    const int64_t run = d.dims[0];
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; idx < d.total; idx += stride) {
+        int64_t o = idx / run;
+        const int64_t i = idx - o * run;
+        int64_t so = d.soff + i, doo = d.doff + i;
+        for (int a = 1; a < d.nd; a++) {
+            const int64_t j = o % d.dims[a];
+            o /= d.dims[a];
+            so += j * d.sstr[a];
+            doo += j * d.dstr[a];
+        }
+        dst[doo] = src[so];
+    }
+}
+
+/* Generic gather/scatter (rare fallback: no contiguous axis on either side
+ * after normalization, e.g. degenerate unit-extent windows). */
+template <typename T>
+__global__ __launch_bounds__(256) void k_copy_generic(const T *__restrict__ src,
+                                                      T *__restrict__ dst,
+                                                      DescDev d)
+{
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; idx < d.total; idx += stride) {
+        int64_t rem = idx, so = d.soff, doo = d.doff;
+        for (int a = 0; a < d.nd; a++) {
+            const int64_t j = rem % d.dims[a];
+            rem /= d.dims[a];
+            so += j * d.sstr[a];
+            doo += j * d.dstr[a];
+        }
+        dst[doo] = src[so];
+    }
+}
+
+/* LDS-tiled N-d transpose: src contiguous along axis 0, dst contiguous along
+ * axis TA.  Reads coalesced along axis 0 into a padded LDS tile, barrier,
+ * writes coalesced along axis TA.  TILE×TILE elements per workgroup,
+ * 64×NROWS threads (wave64-shaped).  Remaining axes are batch.
+ *
+ * The unpack of a permuted pencil (copy_permuted!, Transpositions.jl:588-667)
+ * is exactly this kernel; the reference's GPU path does a generic
+ * permutedims! plus an extra temporary copy (:651-667 "TODO avoid
+ * allocation") — here it is one kernel, no temporary. */
+template <typename T, int TILE, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
+    const T *__restrict__ src, T *__restrict__ dst, DescDev d, int ta,
+    int64_t ntile_i, int64_t ntile_j)
+{
+    __shared__ T tile[TILE][TILE + 1];
+
+    const int tx = threadIdx.x; /* 0..63  : fast axis */
+    const int ty = threadIdx.y; /* 0..NROWS-1 */
+
+    const int64_t tiles_per_batch = ntile_i * ntile_j;
+    const int64_t nbatch_tiles = (int64_t)gridDim.x;
+    (void)nbatch_tiles;
+
+    int64_t bid = blockIdx.x;
+    const int64_t tij = bid % tiles_per_batch;
+    int64_t batch = bid / tiles_per_batch;
+    const int64_t t_i = tij % ntile_i;
+    const int64_t t_j = tij / ntile_i;
+
+    /* batch offsets over axes != 0, != ta */
+    int64_t so_b = d.soff, do_b = d.doff;
+    for (int a = 1; a < d.nd; a++) {
+        if (a == ta) continue;
+        const int64_t j = batch % d.dims[a];
+        batch /= d.dims[a];
+        so_b += j * d.sstr[a];
+        do_b += j * d.dstr[a];
+    }
+
+    const int64_t i0 = t_i * TILE;       /* along axis 0  */
+    const int64_t j0 = t_j * TILE;       /* along axis ta */
+    const int64_t ni = d.dims[0] - i0 < TILE ? d.dims[0] - i0 : TILE;
+    const int64_t nj = d.dims[ta] - j0 < TILE ? d.dims[ta] - j0 : TILE;
+
+    /* load: lanes sweep axis 0 (src-contiguous), rows sweep axis ta */
+    {
+        const int64_t base = so_b + i0 /* *1 */ + j0 * d.sstr[ta];
+        for (int j = ty; j < nj; j += NROWS) {
+            const int64_t row = base + (int64_t)j * d.sstr[ta];
+            for (int i = tx; i < ni; i += 64)
+                tile[j][i] = src[row + i];
+        }
+    }
+    __syncthreads();
+    /* store: lanes sweep axis ta (dst-contiguous), rows sweep axis 0 */
+    {
+        const int64_t base = do_b + j0 /* *1 */ + i0 * d.dstr[0];
+        for (int i = ty; i < ni; i += NROWS) {
+            const int64_t row = base + (int64_t)i * d.dstr[0];
+            for (int j = tx; j < nj; j += 64)
+                dst[row + j] = tile[j][i];
+        }
+    }
+}
+
+/* ================= descriptor normalization & dispatch ============= */
+
+struct CopyDescH {
+    int nd = 0;
+    int64_t dims[MAXND] = {0}, sstr[MAXND] = {0}, dstr[MAXND] = {0};
+    int64_t soff = 0, doff = 0;
+    int64_t total = 0;
+};
+
+static CopyDescH normalize_desc(int nd, const int64_t *dims,
+                                const int64_t *sstr, int64_t soff,
+                                const int64_t *dstr, int64_t doff)
+{
+    struct Axis { int64_t dim, ss, ds; };
+    std::vector<Axis> ax;
+    for (int i = 0; i < nd; i++)
+        if (dims[i] != 1) ax.push_back({dims[i], sstr[i], dstr[i]});
+    if (ax.empty()) ax.push_back({1, 1, 1});
+    std::sort(ax.begin(), ax.end(),
+              [](const Axis &a, const Axis &b) { return a.ss < b.ss; });
+    std::vector<Axis> m{ax[0]};
+    for (size_t i = 1; i < ax.size(); i++) {
+        Axis &p = m.back();
+        if (ax[i].ss == p.ss * p.dim && ax[i].ds == p.ds * p.dim)
+            p.dim *= ax[i].dim;
+        else
+            m.push_back(ax[i]);
+    }
+    CopyDescH out;
+    out.nd = (int)m.size();
+    out.total = 1;
+    for (int i = 0; i < out.nd; i++) {
+        out.dims[i] = m[i].dim;
+        out.sstr[i] = m[i].ss;
+        out.dstr[i] = m[i].ds;
+        out.total *= m[i].dim;
+    }
+    out.soff = soff;
+    out.doff = doff;
+    return out;
+}
+
+/* Try to reinterpret a linear-runs descriptor in W-byte words (W >= esz). */
+static bool word_scale(const CopyDescH &d, int64_t esz, int64_t W,
+                       CopyDescH *out)
+{
+    if ((d.dims[0] * esz) % W || (d.soff * esz) % W || (d.doff * esz) % W)
+        return false;
+    for (int a = 1; a < d.nd; a++)
+        if ((d.sstr[a] * esz) % W || (d.dstr[a] * esz) % W) return false;
+    *out = d;
+    out->dims[0] = d.dims[0] * esz / W;
+    out->soff = d.soff * esz / W;
+    out->doff = d.doff * esz / W;
+    for (int a = 1; a < d.nd; a++) {
+        out->sstr[a] = d.sstr[a] * esz / W;
+        out->dstr[a] = d.dstr[a] * esz / W;
+    }
+    out->total = out->dims[0];
+    for (int a = 1; a < out->nd; a++) out->total *= out->dims[a];
+    return true;
+}
+
+static DescDev to_dev(const CopyDescH &d)
+{
+    DescDev o;
+    o.nd = d.nd;
+    for (int i = 0; i < MAXND; i++) {
+        o.dims[i] = i < d.nd ? d.dims[i] : 1;
+        o.sstr[i] = i < d.nd ? d.sstr[i] : 0;
+        o.dstr[i] = i < d.nd ? d.dstr[i] : 0;
+    }
+    o.soff = d.soff;
+    o.doff = d.doff;
+    o.total = d.total;
+    return o;
+}
+
+static int grid_for(int64_t work_items, int per_block)
+{
+    int64_t blocks = (work_items + per_block - 1) / per_block;
+    /* cap and grid-stride: >> 256 workgroups fills the 8 XCDs; cap keeps
+     * launch latency flat (guide: G11). */
+    const int64_t cap = 2048;
+    if (blocks > cap) blocks = cap;
+    if (blocks < 1) blocks = 1;
+    return (int)blocks;
+}
+
+static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
+                             const void *src, void *dst, hipStream_t stream)
+{
+    if (dn.total == 0) return 0;
+    const char *s = (const char *)src;
+    char *d = (char *)dst;
+
+    const bool lin = (dn.sstr[0] == 1 && dn.dstr[0] == 1);
+    if (lin) {
+        CopyDescH w;
+        int64_t W = 0;
+        for (int64_t cand : {16, 8, 4})
+            if (cand >= esz || (esz % cand) == 0)
+                if (word_scale(dn, esz, cand, &w)) { W = cand; break; }
+        if (!W) { /* odd element size: fall back to bytes */
+            if (!word_scale(dn, esz, 1, &w)) return fail("word_scale(1)");
+            W = 1;
+        }
+        if (w.nd == 1) {
+            const int blocks = grid_for(w.total, 256);
+            if (W == 16)
+                hipLaunchKernelGGL(k_copy_1d<uint4>, dim3(blocks), dim3(256),
+                                   0, stream, (const uint4 *)s + w.soff,
+                                   (uint4 *)d + w.doff, w.total);
+            else if (W == 8)
+                hipLaunchKernelGGL(k_copy_1d<uint64_t>, dim3(blocks),
+                                   dim3(256), 0, stream,
+                                   (const uint64_t *)s + w.soff,
+                                   (uint64_t *)d + w.doff, w.total);
+            else if (W == 4)
+                hipLaunchKernelGGL(k_copy_1d<uint32_t>, dim3(blocks),
+                                   dim3(256), 0, stream,
+                                   (const uint32_t *)s + w.soff,
+                                   (uint32_t *)d + w.doff, w.total);
+            else
+                hipLaunchKernelGGL(k_copy_1d<uint8_t>, dim3(blocks), dim3(256),
+                                   0, stream, (const uint8_t *)s + w.soff,
+                                   (uint8_t *)d + w.doff, w.total);
+        } else {
+            DescDev dd = to_dev(w);
+            const int blocks = grid_for(w.total, 256);
+            if (W == 16)
+                hipLaunchKernelGGL(k_copy_linear<uint4>, dim3(blocks),
+                                   dim3(256), 0, stream, (const uint4 *)s,
+                                   (uint4 *)d, dd);
+            else if (W == 8)
+                hipLaunchKernelGGL(k_copy_linear<uint64_t>, dim3(blocks),
+                                   dim3(256), 0, stream, (const uint64_t *)s,
+                                   (uint64_t *)d, dd);
+            else if (W == 4)
+                hipLaunchKernelGGL(k_copy_linear<uint32_t>, dim3(blocks),
+                                   dim3(256), 0, stream, (const uint32_t *)s,
+                                   (uint32_t *)d, dd);
+            else
+                hipLaunchKernelGGL(k_copy_linear<uint8_t>, dim3(blocks),
+                                   dim3(256), 0, stream, (const uint8_t *)s,
+                                   (uint8_t *)d, dd);
+        }
+        HIP_CHECK(hipGetLastError());
+        return 0;
+    }
+
+    /* transpose case: src-contiguous axis 0, find dst-contiguous axis */
+    int ta = -1;
+    if (dn.sstr[0] == 1)
+        for (int a = 1; a < dn.nd; a++)
+            if (dn.dstr[a] == 1) { ta = a; break; }
+
+    if (ta > 0 && (esz == 4 || esz == 8 || esz == 16)) {
+        DescDev dd = to_dev(dn);
+        constexpr int TILE = 64, NROWS = 8;
+        const int64_t nti = (dn.dims[0] + TILE - 1) / TILE;
+        const int64_t ntj = (dn.dims[ta] + TILE - 1) / TILE;
+        int64_t nbatch = 1;
+        for (int a = 1; a < dn.nd; a++)
+            if (a != ta) nbatch *= dn.dims[a];
+        const int64_t blocks = nti * ntj * nbatch;
+        if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
+        if (esz == 8)
+            hipLaunchKernelGGL((k_transpose_tile<uint64_t, TILE, NROWS>),
+                               dim3((uint32_t)blocks), dim3(64, NROWS), 0,
+                               stream, (const uint64_t *)s, (uint64_t *)d, dd,
+                               ta, nti, ntj);
+        else if (esz == 4)
+            hipLaunchKernelGGL((k_transpose_tile<uint32_t, TILE, NROWS>),
+                               dim3((uint32_t)blocks), dim3(64, NROWS), 0,
+                               stream, (const uint32_t *)s, (uint32_t *)d, dd,
+                               ta, nti, ntj);
+        else {
+            constexpr int TILE16 = 32; /* 16-B elements: 32×32 = 16 KiB+pad */
+            const int64_t nti2 = (dn.dims[0] + TILE16 - 1) / TILE16;
+            const int64_t ntj2 = (dn.dims[ta] + TILE16 - 1) / TILE16;
+            const int64_t blocks2 = nti2 * ntj2 * nbatch;
+            hipLaunchKernelGGL((k_transpose_tile<uint4, TILE16, NROWS>),
+                               dim3((uint32_t)blocks2), dim3(64, NROWS), 0,
+                               stream, (const uint4 *)s, (uint4 *)d, dd, ta,
+                               nti2, ntj2);
+        }
+        HIP_CHECK(hipGetLastError());
+        return 0;
+    }
+
+    /* generic fallback */
+    {
+        DescDev dd = to_dev(dn);
+        const int blocks = grid_for(dn.total, 256);
+        if (esz == 16)
+            hipLaunchKernelGGL(k_copy_generic<uint4>, dim3(blocks), dim3(256),
+                               0, stream, (const uint4 *)s, (uint4 *)d, dd);
+        else if (esz == 8)
+            hipLaunchKernelGGL(k_copy_generic<uint64_t>, dim3(blocks),
+                               dim3(256), 0, stream, (const uint64_t *)s,
+                               (uint64_t *)d, dd);
+        else if (esz == 4)
+            hipLaunchKernelGGL(k_copy_generic<uint32_t>, dim3(blocks),
+                               dim3(256), 0, stream, (const uint32_t *)s,
+                               (uint32_t *)d, dd);
+        else
+            return fail("unsupported element size %lld", (long long)esz);
+        HIP_CHECK(hipGetLastError());
+        return 0;
+    }
+}
+
+/* ================= metadata ======================================== */
+
+struct Range { int64_t lo, hi; };
+
+struct pa_topology {
+    int m;
+    std::vector<int64_t> dims;
+    int64_t nranks;
+};
+
+struct pa_pencil {
+    pa_topology topo; /* copied: a pencil owns its grid description */
+    int n;
+    std::vector<int64_t> size_global;
+    std::vector<int32_t> decomp; /* length m */
+    std::vector<int32_t> perm;   /* length n; perm[i] = logical dim at mem i */
+};
+
+static void split_range(int64_t c, int64_t P, int64_t N, Range *r)
+{
+    r->lo = (N * c) / P; /* data_ranges.jl:4-9 */
+    r->hi = (N * (c + 1)) / P;
+}
+
+static void cart_coords(const pa_topology &t, int rank, int64_t *coords)
+{
+    int64_t rem = rank;
+    for (int i = 0; i < t.m; i++) {
+        int64_t stride = 1;
+        for (int j = i + 1; j < t.m; j++) stride *= t.dims[j];
+        coords[i] = rem / stride;
+        rem %= stride;
+    }
+}
+
+static int cart_rank(const pa_topology &t, const int64_t *coords)
+{
+    int64_t r = 0;
+    for (int i = 0; i < t.m; i++) r = r * t.dims[i] + coords[i];
+    return (int)r;
+}
+
+static void axes_for_coords(const pa_pencil &p, const int64_t *coords,
+                            Range *out)
+{
+    for (int d = 0; d < p.n; d++) {
+        out[d].lo = 0;
+        out[d].hi = p.size_global[d];
+    }
+    for (int j = 0; j < p.topo.m; j++)
+        split_range(coords[j], p.topo.dims[j], p.size_global[p.decomp[j]],
+                    &out[p.decomp[j]]);
+}
+
+static void axes_for_rank(const pa_pencil &p, int rank, Range *out)
+{
+    int64_t coords[MAXND];
+    cart_coords(p.topo, rank, coords);
+    axes_for_coords(p, coords, out);
+}
+
+static void range_intersect(const Range &a, const Range &b, Range *out)
+{
+    out->lo = std::max(a.lo, b.lo);
+    out->hi = std::max(out->lo, std::min(a.hi, b.hi));
+}
+
+static int64_t region_nelem(int n, const Range *r)
+{
+    int64_t v = 1;
+    for (int d = 0; d < n; d++) v *= (r[d].hi - r[d].lo);
+    return v;
+}
+
+/* ================= comm ============================================ */
+
+struct pa_comm {
+    ncclComm_t comm;
+    int nranks, rank;
+};
+
+/* ================= plan ============================================ */
+
+struct PeerBlockC {
+    int k, grank;
+    int64_t send_off, recv_off, send_n, recv_n;
+    bool has_pack = false, has_unpack = false;
+    CopyDescH pack, unpack;
+};
+
+struct pa_plan {
+    pa_pencil Pi, Po;
+    int rank;
+    int E;
+    std::vector<int64_t> extra;
+    int64_t esz;
+    int R;    /* -1 = same decomposition */
+    int P;    /* subgroup size */
+    int myk;  /* my coordinate along R */
+    std::vector<PeerBlockC> peers;
+    bool has_local = false;
+    CopyDescH local;
+    int64_t send_total = 0, recv_total = 0; /* elements (remote blocks) */
+    void *send_buf = nullptr, *recv_buf = nullptr;
+    bool own_bufs = false;
+    pa_comm *comm = nullptr;
+};
+
+static int64_t prod_extra(const pa_plan &pl)
+{
+    int64_t v = 1;
+    for (auto e : pl.extra) v *= e;
+    return v;
+}
+
+/* parent memory dims (memory-order local sizes + extra) and column-major
+ * strides (axis 0 fastest — the Julia parent, arrays.jl:134-138). */
+static void parent_strides(const pa_pencil &p, int rank,
+                           const std::vector<int64_t> &extra, int64_t *mem,
+                           int64_t *st, int *knd)
+{
+    Range ax[MAXND];
+    axes_for_rank(p, rank, ax);
+    int k = 0;
+    for (int i = 0; i < p.n; i++, k++)
+        mem[k] = ax[p.perm[i]].hi - ax[p.perm[i]].lo;
+    for (auto e : extra) mem[k++] = e;
+    int64_t acc = 1;
+    for (int i = 0; i < k; i++) {
+        st[i] = acc;
+        acc *= mem[i];
+    }
+    *knd = k;
+}
+
+/* window of a global region in the parent of pencil p (axes in p's memory
+ * order + extras): dims/strides/offset (to_local, Pencils.jl:579-587). */
+static void window_desc(const pa_pencil &p, int rank,
+                        const std::vector<int64_t> &extra, const Range *region,
+                        int64_t *dims, int64_t *str, int64_t *off, int *knd)
+{
+    Range ax[MAXND];
+    axes_for_rank(p, rank, ax);
+    int64_t mem[2 * MAXND], pst[2 * MAXND];
+    int k;
+    parent_strides(p, rank, extra, mem, pst, &k);
+    int64_t o = 0;
+    for (int i = 0; i < p.n; i++) {
+        const int d = p.perm[i];
+        dims[i] = region[d].hi - region[d].lo;
+        o += (region[d].lo - ax[d].lo) * pst[i];
+    }
+    for (size_t e = 0; e < extra.size(); e++) dims[p.n + e] = extra[e];
+    for (int i = 0; i < k; i++) str[i] = pst[i];
+    *off = o;
+    *knd = k;
+}
+
+/* buffer->dest unpack descriptor: buffer axes = Pi memory order (+extras),
+ * column-major with given strides; dest axis i' reads buffer axis
+ * inv(perm_i)[perm_o[i']] (the relative permutation, Transpositions.jl:506).
+ */
+static CopyDescH unpack_desc(const pa_plan &pl, const Range *grange,
+                             const int64_t *bufdims, const int64_t *bufstr,
+                             int64_t bufoff)
+{
+    const pa_pencil &Po = pl.Po;
+    const pa_pencil &Pi = pl.Pi;
+    const int n = Pi.n;
+    const int E = pl.E;
+
+    Range axo[MAXND];
+    axes_for_rank(Po, pl.rank, axo);
+    int64_t mem_o[2 * MAXND], pst_o[2 * MAXND];
+    int k;
+    parent_strides(Po, pl.rank, pl.extra, mem_o, pst_o, &k);
+
+    int32_t ipi[MAXND];
+    for (int i = 0; i < n; i++) ipi[Pi.perm[i]] = i;
+
+    int64_t dstr[2 * MAXND], doff = 0;
+    for (int ip = 0; ip < n; ip++) {
+        const int d = Po.perm[ip];
+        dstr[ipi[d]] = pst_o[ip];
+        doff += (grange[d].lo - axo[d].lo) * pst_o[ip];
+    }
+    for (int e = 0; e < E; e++) dstr[n + e] = pst_o[n + e];
+
+    return normalize_desc(n + E, bufdims, bufstr, bufoff, dstr, doff);
+}
+
+extern "C" {
+
+/* ---- topology ----------------------------------------------------- */
+
+pa_status pa_topology_create(int m, const int64_t *pdims, pa_topology **out)
+{
+    if (m <= 0 || m > MAXND) return fail("invalid topology ndims %d", m);
+    auto *t = new pa_topology;
+    t->m = m;
+    t->nranks = 1;
+    for (int i = 0; i < m; i++) {
+        if (pdims[i] <= 0) { delete t; return fail("invalid pdims"); }
+        t->dims.push_back(pdims[i]);
+        t->nranks *= pdims[i];
+    }
+    *out = t;
+    return 0;
+}
+
+void pa_topology_destroy(pa_topology *t) { delete t; }
+int pa_topology_nranks(const pa_topology *t) { return (int)t->nranks; }
+
+pa_status pa_topology_coords(const pa_topology *t, int rank, int64_t *coords)
+{
+    if (rank < 0 || rank >= t->nranks) return fail("rank out of range");
+    cart_coords(*t, rank, coords);
+    return 0;
+}
+
+/* ---- pencil ------------------------------------------------------- */
+
+pa_status pa_pencil_create(pa_topology *t, int n, const int64_t *size_global,
+                           const int32_t *decomp_dims, const int32_t *perm,
+                           pa_pencil **out)
+{
+    if (n <= 0 || n + 2 > MAXND) return fail("invalid ndims %d", n);
+    if (t->m > n) return fail("M (%d) cannot exceed N (%d)", t->m, n);
+    auto *p = new pa_pencil;
+    p->topo = *t;
+    p->n = n;
+    p->size_global.assign(size_global, size_global + n);
+    p->decomp.assign(decomp_dims, decomp_dims + t->m);
+    /* _check_selected_dimensions (Pencils.jl:393-406) */
+    for (int j = 0; j < t->m; j++) {
+        if (p->decomp[j] < 0 || p->decomp[j] >= n) {
+            delete p;
+            return fail("decomp dim %d out of range", p->decomp[j]);
+        }
+        for (int j2 = 0; j2 < j; j2++)
+            if (p->decomp[j2] == p->decomp[j]) {
+                delete p;
+                return fail("repeated decomp dim %d", p->decomp[j]);
+            }
+    }
+    if (perm) {
+        p->perm.assign(perm, perm + n);
+        std::vector<int> seen(n, 0);
+        for (int i = 0; i < n; i++) {
+            if (p->perm[i] < 0 || p->perm[i] >= n || seen[p->perm[i]]++) {
+                delete p;
+                return fail("invalid permutation");
+            }
+        }
+    } else {
+        for (int i = 0; i < n; i++) p->perm.push_back(i);
+    }
+    *out = p;
+    return 0;
+}
+
+void pa_pencil_destroy(pa_pencil *p) { delete p; }
+
+pa_status pa_pencil_range_local(const pa_pencil *p, int rank, int memory_order,
+                                int64_t *lo, int64_t *hi)
+{
+    if (rank < 0 || rank >= p->topo.nranks) return fail("rank out of range");
+    Range ax[MAXND];
+    axes_for_rank(*p, rank, ax);
+    for (int i = 0; i < p->n; i++) {
+        const int d = memory_order ? p->perm[i] : i;
+        lo[i] = ax[d].lo;
+        hi[i] = ax[d].hi;
+    }
+    return 0;
+}
+
+pa_status pa_pencil_size_local(const pa_pencil *p, int rank, int memory_order,
+                               int64_t *out)
+{
+    int64_t lo[MAXND], hi[MAXND];
+    pa_status st = pa_pencil_range_local(p, rank, memory_order, lo, hi);
+    if (st) return st;
+    for (int i = 0; i < p->n; i++) out[i] = hi[i] - lo[i];
+    return 0;
+}
+
+int64_t pa_pencil_length_local(const pa_pencil *p, int rank)
+{
+    Range ax[MAXND];
+    axes_for_rank(*p, rank, ax);
+    return region_nelem(p->n, ax);
+}
+
+/* ---- comm --------------------------------------------------------- */
+
+int pa_unique_id_size(void) { return (int)sizeof(ncclUniqueId); }
+
+pa_status pa_get_unique_id(char *id)
+{
+    NCCL_CHECK(ncclGetUniqueId((ncclUniqueId *)id));
+    return 0;
+}
+
+pa_status pa_comm_create(const char *id, int nranks, int rank, pa_comm **out)
+{
+    auto *c = new pa_comm;
+    c->nranks = nranks;
+    c->rank = rank;
+    ncclUniqueId uid;
+    memcpy(&uid, id, sizeof uid);
+    ncclResult_t r = ncclCommInitRank(&c->comm, nranks, uid, rank);
+    if (r != ncclSuccess) {
+        delete c;
+        return fail("ncclCommInitRank: %s", ncclGetErrorString(r));
+    }
+    *out = c;
+    return 0;
+}
+
+void pa_comm_destroy(pa_comm *c)
+{
+    if (!c) return;
+    ncclCommDestroy(c->comm);
+    delete c;
+}
+
+/* ---- plan --------------------------------------------------------- */
+
+pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
+                         int64_t elem_size, int e, const int64_t *extra_dims,
+                         int rank, pa_plan **out)
+{
+    /* assert_compatible (Transpositions.jl:182-199) */
+    if (pin->topo.m != pout->topo.m || pin->topo.dims != pout->topo.dims)
+        return fail("pencil topologies must be the same.");
+    if (pin->n != pout->n || pin->size_global != pout->size_global)
+        return fail("global data sizes must be the same between different "
+                    "pencil configurations.");
+    int ndiff = 0, R = -1;
+    for (int j = 0; j < pin->topo.m; j++)
+        if (pin->decomp[j] != pout->decomp[j]) {
+            ndiff++;
+            if (R < 0) R = j;
+        }
+    if (ndiff > 1)
+        return fail("pencil decompositions must differ in at most one "
+                    "dimension.");
+    if (rank < 0 || rank >= pin->topo.nranks) return fail("rank out of range");
+    if (elem_size <= 0) return fail("invalid elem_size");
+    if (pin->n + e + 1 > 2 * MAXND) return fail("too many dims");
+
+    auto *pl = new pa_plan;
+    pl->Pi = *pin;
+    pl->Po = *pout;
+    pl->rank = rank;
+    pl->E = e;
+    if (e) pl->extra.assign(extra_dims, extra_dims + e);
+    pl->esz = elem_size;
+    pl->R = R;
+
+    const int n = pin->n;
+    const int64_t pex = prod_extra(*pl);
+
+    if (R < 0) {
+        pl->P = 1;
+        pl->myk = 0;
+        /* local path (transpose_impl!(::Nothing), Transpositions.jl:214-271):
+         * one fused copy/permute */
+        Range region[MAXND];
+        axes_for_rank(*pin, rank, region);
+        if (region_nelem(n, region) > 0) {
+            int64_t dims[2 * MAXND], str[2 * MAXND], off;
+            int k;
+            window_desc(*pin, rank, pl->extra, region, dims, str, &off, &k);
+            pl->local = unpack_desc(*pl, region, dims, str, off);
+            pl->has_local = true;
+        }
+        *out = pl;
+        return 0;
+    }
+
+    pl->P = (int)pin->topo.dims[R];
+    int64_t coords[MAXND];
+    cart_coords(pin->topo, rank, coords);
+    pl->myk = (int)coords[R];
+
+    Range axl_i[MAXND], axl_o[MAXND];
+    axes_for_rank(*pin, rank, axl_i);
+    axes_for_rank(*pout, rank, axl_o);
+
+    int64_t isend = 0, irecv = 0;
+    for (int k = 0; k < pl->P; k++) {
+        int64_t pc[MAXND];
+        memcpy(pc, coords, sizeof(int64_t) * pin->topo.m);
+        pc[R] = k;
+        PeerBlockC blk;
+        blk.k = k;
+        blk.grank = cart_rank(pin->topo, pc);
+
+        Range axp_o[MAXND], axp_i[MAXND], sr[MAXND], rr[MAXND];
+        axes_for_coords(*pout, pc, axp_o);
+        axes_for_coords(*pin, pc, axp_i);
+        for (int d = 0; d < n; d++) {
+            range_intersect(axl_i[d], axp_o[d], &sr[d]); /* :383 */
+            range_intersect(axl_o[d], axp_i[d], &rr[d]); /* :388,:521 */
+        }
+        blk.send_n = region_nelem(n, sr) * pex;
+        blk.recv_n = region_nelem(n, rr) * pex;
+
+        if (k == pl->myk) {
+            blk.send_off = 0;
+            blk.recv_off = 0;
+            /* fused self path: direct window->window permuted copy
+             * (replaces :394-404 + :588-606; same values, half the HBM
+             * traffic) */
+            if (blk.send_n > 0) {
+                int64_t dims[2 * MAXND], str[2 * MAXND], off;
+                int kk;
+                window_desc(*pin, rank, pl->extra, sr, dims, str, &off, &kk);
+                pl->local = unpack_desc(*pl, rr, dims, str, off);
+                pl->has_local = true;
+            }
+        } else {
+            blk.send_off = isend;
+            blk.recv_off = irecv;
+            if (blk.send_n > 0) {
+                int64_t dims[2 * MAXND], str[2 * MAXND], off;
+                int kk;
+                window_desc(*pin, rank, pl->extra, sr, dims, str, &off, &kk);
+                int64_t cstr[2 * MAXND], acc = 1;
+                for (int i = 0; i < kk; i++) {
+                    cstr[i] = acc;
+                    acc *= dims[i];
+                }
+                blk.pack = normalize_desc(kk, dims, str, off, cstr, isend);
+                blk.has_pack = true;
+            }
+            if (blk.recv_n > 0) {
+                /* buffer dims: recv block extents gathered by Pi's
+                 * permutation (+ extras), column-major (:527,:599-600) */
+                int64_t bdims[2 * MAXND];
+                for (int i = 0; i < n; i++)
+                    bdims[i] = rr[pin->perm[i]].hi - rr[pin->perm[i]].lo;
+                for (int e2 = 0; e2 < e; e2++) bdims[n + e2] = pl->extra[e2];
+                int64_t bstr[2 * MAXND], acc = 1;
+                for (int i = 0; i < n + e; i++) {
+                    bstr[i] = acc;
+                    acc *= bdims[i];
+                }
+                blk.unpack = unpack_desc(*pl, rr, bdims, bstr, irecv);
+                blk.has_unpack = true;
+            }
+            isend += blk.send_n;
+            irecv += blk.recv_n;
+        }
+        pl->peers.push_back(blk);
+    }
+    pl->send_total = isend;
+    pl->recv_total = irecv;
+    *out = pl;
+    return 0;
+}
+
+void pa_plan_destroy(pa_plan *p)
+{
+    if (!p) return;
+    if (p->own_bufs) {
+        if (p->send_buf) (void)hipFree(p->send_buf);
+        if (p->recv_buf) (void)hipFree(p->recv_buf);
+    }
+    delete p;
+}
+
+pa_status pa_plan_set_comm(pa_plan *p, pa_comm *c)
+{
+    if (p->R >= 0 && p->P > 1) {
+        if (c->nranks != p->P)
+            return fail("comm size %d != subgroup size %d", c->nranks, p->P);
+        if (c->rank != p->myk)
+            return fail("comm rank %d != my subgroup coordinate %d", c->rank,
+                        p->myk);
+    }
+    p->comm = c;
+    return 0;
+}
+
+pa_status pa_plan_buffer_sizes(const pa_plan *p, int64_t *send_bytes,
+                               int64_t *recv_bytes)
+{
+    *send_bytes = p->send_total * p->esz;
+    *recv_bytes = p->recv_total * p->esz;
+    return 0;
+}
+
+pa_status pa_plan_set_buffers(pa_plan *p, void *send_buf, void *recv_buf)
+{
+    if (p->own_bufs) {
+        if (p->send_buf) (void)hipFree(p->send_buf);
+        if (p->recv_buf) (void)hipFree(p->recv_buf);
+        p->own_bufs = false;
+    }
+    p->send_buf = send_buf;
+    p->recv_buf = recv_buf;
+    return 0;
+}
+
+pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
+                               void *dst_parent, void *stream_)
+{
+    hipStream_t stream = (hipStream_t)stream_;
+
+    const bool need_bufs = p->send_total > 0 || p->recv_total > 0;
+    if (need_bufs && !p->send_buf && !p->own_bufs) {
+        if (p->send_total)
+            HIP_CHECK(hipMalloc(&p->send_buf, p->send_total * p->esz));
+        if (p->recv_total)
+            HIP_CHECK(hipMalloc(&p->recv_buf, p->recv_total * p->esz));
+        p->own_bufs = true;
+    }
+
+    /* 1. pack every remote block (Transpositions.jl:346-431) */
+    for (auto &blk : p->peers)
+        if (blk.has_pack) {
+            pa_status st =
+                launch_desc(blk.pack, p->esz, src_parent, p->send_buf, stream);
+            if (st) return st;
+        }
+
+    /* 2. exchange: grouped ncclSend/ncclRecv over xGMI, stream-ordered
+     * after the pack kernels (replaces :419-428/:463-479; the pre-send
+     * device sync of :472-473 is unnecessary — RCCL is stream-ordered). */
+    if (p->R >= 0 && p->P > 1) {
+        bool any = false;
+        for (auto &blk : p->peers)
+            if (blk.k != p->myk && (blk.send_n || blk.recv_n)) any = true;
+        if (any) {
+            if (!p->comm)
+                return fail("subgroup exchange requires pa_plan_set_comm");
+            NCCL_CHECK(ncclGroupStart());
+            for (auto &blk : p->peers) {
+                if (blk.k == p->myk) continue;
+                if (blk.recv_n)
+                    NCCL_CHECK(ncclRecv((char *)p->recv_buf +
+                                            blk.recv_off * p->esz,
+                                        (size_t)(blk.recv_n * p->esz),
+                                        ncclUint8, blk.k, p->comm->comm,
+                                        stream));
+                if (blk.send_n)
+                    NCCL_CHECK(ncclSend((const char *)p->send_buf +
+                                            blk.send_off * p->esz,
+                                        (size_t)(blk.send_n * p->esz),
+                                        ncclUint8, blk.k, p->comm->comm,
+                                        stream));
+            }
+            NCCL_CHECK(ncclGroupEnd());
+        }
+    }
+
+    /* 3. fused local/self copy (independent of the exchange) */
+    if (p->has_local) {
+        pa_status st =
+            launch_desc(p->local, p->esz, src_parent, dst_parent, stream);
+        if (st) return st;
+    }
+
+    /* 4. unpack every received block (:489-536) */
+    for (auto &blk : p->peers)
+        if (blk.has_unpack) {
+            pa_status st = launch_desc(blk.unpack, p->esz, p->recv_buf,
+                                       dst_parent, stream);
+            if (st) return st;
+        }
+
+    return 0;
+}
+
+pa_status pa_transpose_wait(pa_plan *p, void *stream_)
+{
+    (void)p;
+    HIP_CHECK(hipStreamSynchronize((hipStream_t)stream_));
+    return 0;
+}
+
+/* ---- introspection ------------------------------------------------ */
+
+int pa_plan_nproc_sub(const pa_plan *p) { return p->P; }
+int pa_plan_r_dim(const pa_plan *p) { return p->R; }
+int pa_plan_my_k(const pa_plan *p) { return p->myk; }
+
+pa_status pa_plan_block_info(const pa_plan *p, int k, int64_t out[8])
+{
+    if (p->R < 0 || k < 0 || k >= (int)p->peers.size())
+        return fail("no peer block %d", k);
+    const PeerBlockC &b = p->peers[k];
+    out[0] = b.k;
+    out[1] = b.grank;
+    out[2] = b.send_off;
+    out[3] = b.recv_off;
+    out[4] = b.send_n;
+    out[5] = b.recv_n;
+    out[6] = b.has_pack;
+    out[7] = b.has_unpack;
+    return 0;
+}
+
+pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,
+                           int64_t *dims, int64_t *sstr, int64_t *soff,
+                           int64_t *dstr, int64_t *doff)
+{
+    const CopyDescH *d = nullptr;
+    if (which == 0) {
+        if (!p->has_local) return fail("no local copy in plan");
+        d = &p->local;
+    } else {
+        if (p->R < 0 || k < 0 || k >= (int)p->peers.size())
+            return fail("no peer block %d", k);
+        const PeerBlockC &b = p->peers[k];
+        if (which == 1) {
+            if (!b.has_pack) return fail("no pack for peer %d", k);
+            d = &b.pack;
+        } else if (which == 2) {
+            if (!b.has_unpack) return fail("no unpack for peer %d", k);
+            d = &b.unpack;
+        } else
+            return fail("bad which %d", which);
+    }
+    *nd = d->nd;
+    for (int i = 0; i < d->nd; i++) {
+        dims[i] = d->dims[i];
+        sstr[i] = d->sstr[i];
+        dstr[i] = d->dstr[i];
+    }
+    *soff = d->soff;
+    *doff = d->doff;
+    return 0;
+}
+
+/* ---- standalone device copy --------------------------------------- */
+
+pa_status pa_device_copy(int nd, const int64_t *dims, const int64_t *sstr,
+                         int64_t soff, const int64_t *dstr, int64_t doff,
+                         int64_t elem_size, const void *src, void *dst,
+                         void *stream)
+{
+    if (nd <= 0 || nd > MAXND) return fail("bad nd");
+    CopyDescH d = normalize_desc(nd, dims, sstr, soff, dstr, doff);
+    return launch_desc(d, elem_size, src, dst, (hipStream_t)stream);
+}
+
+} /* extern "C" */
