@@ -527,6 +527,12 @@ struct pa_plan {
     void *send_buf = nullptr, *recv_buf = nullptr;
     bool own_bufs = false;
     pa_comm *comm = nullptr;
+    /* two-stream overlap: the exchange runs on comm_stream while the fused
+     * local copy runs on the caller's stream (the reference overlaps the
+     * local block with the exchange the same way: it copies it first and
+     * unpacks it while recvs are in flight, Transpositions.jl:510-517). */
+    hipStream_t comm_stream = nullptr;
+    hipEvent_t ev_pack = nullptr, ev_comm = nullptr;
 };
 
 static int64_t prod_extra(const pa_plan &pl)
@@ -892,6 +898,9 @@ void pa_plan_destroy(pa_plan *p)
         if (p->send_buf) (void)hipFree(p->send_buf);
         if (p->recv_buf) (void)hipFree(p->recv_buf);
     }
+    if (p->ev_pack) (void)hipEventDestroy(p->ev_pack);
+    if (p->ev_comm) (void)hipEventDestroy(p->ev_comm);
+    if (p->comm_stream) (void)hipStreamDestroy(p->comm_stream);
     delete p;
 }
 
@@ -950,44 +959,60 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
             if (st) return st;
         }
 
-    /* 2. exchange: grouped ncclSend/ncclRecv over xGMI, stream-ordered
-     * after the pack kernels (replaces :419-428/:463-479; the pre-send
-     * device sync of :472-473 is unnecessary — RCCL is stream-ordered). */
+    /* 2. exchange: grouped ncclSend/ncclRecv over xGMI on a dedicated comm
+     * stream ordered after the pack kernels (replaces :419-428/:463-479;
+     * the pre-send device sync of :472-473 is unnecessary — everything is
+     * stream/event-ordered).  The fused local copy (step 3) runs on the
+     * caller's stream CONCURRENTLY with the exchange — the engine's
+     * equivalent of the reference's unpack-local-while-recvs-fly overlap
+     * (Transpositions.jl:510-517). */
+    bool exchanging = false;
     if (p->R >= 0 && p->P > 1) {
-        bool any = false;
         for (auto &blk : p->peers)
-            if (blk.k != p->myk && (blk.send_n || blk.recv_n)) any = true;
-        if (any) {
-            if (!p->comm)
-                return fail("subgroup exchange requires pa_plan_set_comm");
-            NCCL_CHECK(ncclGroupStart());
-            for (auto &blk : p->peers) {
-                if (blk.k == p->myk) continue;
-                if (blk.recv_n)
-                    NCCL_CHECK(ncclRecv((char *)p->recv_buf +
-                                            blk.recv_off * p->esz,
-                                        (size_t)(blk.recv_n * p->esz),
-                                        ncclUint8, blk.k, p->comm->comm,
-                                        stream));
-                if (blk.send_n)
-                    NCCL_CHECK(ncclSend((const char *)p->send_buf +
-                                            blk.send_off * p->esz,
-                                        (size_t)(blk.send_n * p->esz),
-                                        ncclUint8, blk.k, p->comm->comm,
-                                        stream));
-            }
-            NCCL_CHECK(ncclGroupEnd());
+            if (blk.k != p->myk && (blk.send_n || blk.recv_n))
+                exchanging = true;
+    }
+    if (exchanging) {
+        if (!p->comm)
+            return fail("subgroup exchange requires pa_plan_set_comm");
+        if (!p->comm_stream) {
+            HIP_CHECK(hipStreamCreateWithFlags(&p->comm_stream,
+                                               hipStreamNonBlocking));
+            HIP_CHECK(hipEventCreateWithFlags(&p->ev_pack,
+                                              hipEventDisableTiming));
+            HIP_CHECK(hipEventCreateWithFlags(&p->ev_comm,
+                                              hipEventDisableTiming));
         }
+        HIP_CHECK(hipEventRecord(p->ev_pack, stream));
+        HIP_CHECK(hipStreamWaitEvent(p->comm_stream, p->ev_pack, 0));
+        NCCL_CHECK(ncclGroupStart());
+        for (auto &blk : p->peers) {
+            if (blk.k == p->myk) continue;
+            if (blk.recv_n)
+                NCCL_CHECK(ncclRecv((char *)p->recv_buf + blk.recv_off * p->esz,
+                                    (size_t)(blk.recv_n * p->esz), ncclUint8,
+                                    blk.k, p->comm->comm, p->comm_stream));
+            if (blk.send_n)
+                NCCL_CHECK(ncclSend((const char *)p->send_buf +
+                                        blk.send_off * p->esz,
+                                    (size_t)(blk.send_n * p->esz), ncclUint8,
+                                    blk.k, p->comm->comm, p->comm_stream));
+        }
+        NCCL_CHECK(ncclGroupEnd());
+        HIP_CHECK(hipEventRecord(p->ev_comm, p->comm_stream));
     }
 
-    /* 3. fused local/self copy (independent of the exchange) */
+    /* 3. fused local/self copy — on the caller's stream, overlapping the
+     * exchange */
     if (p->has_local) {
         pa_status st =
             launch_desc(p->local, p->esz, src_parent, dst_parent, stream);
         if (st) return st;
     }
 
-    /* 4. unpack every received block (:489-536) */
+    /* 4. unpack every received block (:489-536) after the exchange lands */
+    if (exchanging)
+        HIP_CHECK(hipStreamWaitEvent(stream, p->ev_comm, 0));
     for (auto &blk : p->peers)
         if (blk.has_unpack) {
             pa_status st = launch_desc(blk.unpack, p->esz, p->recv_buf,

@@ -1,0 +1,321 @@
+/* probe_copy.hip — kernel-variant microbenchmark for the copy engine.
+ * Standalone; run on an MI355X to pick the fastest variants, which are then
+ * baked into csrc/pencilhip.hip.  Measures GB/s (algorithmic: rd+wr bytes /
+ * time) with HIP events, best-of-reps, plus a checksum correctness check.
+ *
+ * Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 probe_copy.hip -o probe_copy
+ */
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+
+#define CHK(x)                                                               \
+    do {                                                                     \
+        hipError_t e = (x);                                                  \
+        if (e != hipSuccess) {                                               \
+            fprintf(stderr, "HIP error %s @%d\n", hipGetErrorString(e),      \
+                    __LINE__);                                               \
+            exit(1);                                                         \
+        }                                                                    \
+    } while (0)
+
+/* ---------------- 1-D copy variants ---------------- */
+
+__global__ __launch_bounds__(256) void c_gridstride(const uint4 *__restrict__ s,
+                                                    uint4 *__restrict__ d,
+                                                    int64_t n)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t st = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += st) d[i] = s[i];
+}
+
+__global__ __launch_bounds__(256) void c_direct(const uint4 *__restrict__ s,
+                                                uint4 *__restrict__ d,
+                                                int64_t n)
+{
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) d[i] = s[i];
+}
+
+typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+
+__global__ __launch_bounds__(256) void c_nt(const v4u *__restrict__ s,
+                                            v4u *__restrict__ d, int64_t n)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t st = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += st)
+        __builtin_nontemporal_store(__builtin_nontemporal_load(&s[i]), &d[i]);
+}
+
+__global__ __launch_bounds__(256) void c_unroll4(const uint4 *__restrict__ s,
+                                                 uint4 *__restrict__ d,
+                                                 int64_t n)
+{
+    const int64_t st = (int64_t)gridDim.x * blockDim.x;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t n4 = n - 3 * st;
+    for (; i < n4; i += 4 * st) {
+        uint4 a = s[i], b = s[i + st], c = s[i + 2 * st], e = s[i + 3 * st];
+        d[i] = a;
+        d[i + st] = b;
+        d[i + 2 * st] = c;
+        d[i + 3 * st] = e;
+    }
+    for (; i < n; i += st) d[i] = s[i];
+}
+
+__global__ __launch_bounds__(512) void c_direct512(const uint4 *__restrict__ s,
+                                                   uint4 *__restrict__ d,
+                                                   int64_t n)
+{
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) d[i] = s[i];
+}
+
+__global__ __launch_bounds__(256) void c_nt_unroll4(const v4u *__restrict__ s,
+                                                    v4u *__restrict__ d,
+                                                    int64_t n)
+{
+    const int64_t st = (int64_t)gridDim.x * blockDim.x;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t n4 = n - 3 * st;
+    for (; i < n4; i += 4 * st) {
+        v4u a = __builtin_nontemporal_load(&s[i]);
+        v4u b = __builtin_nontemporal_load(&s[i + st]);
+        v4u c = __builtin_nontemporal_load(&s[i + 2 * st]);
+        v4u e = __builtin_nontemporal_load(&s[i + 3 * st]);
+        __builtin_nontemporal_store(a, &d[i]);
+        __builtin_nontemporal_store(b, &d[i + st]);
+        __builtin_nontemporal_store(c, &d[i + 2 * st]);
+        __builtin_nontemporal_store(e, &d[i + 3 * st]);
+    }
+    for (; i < n; i += st) d[i] = s[i];
+}
+
+/* ---------------- transpose variants ----------------
+ * Model: 2-D transpose of a NI x NJ f64 matrix (src row-major along i,
+ * dst row-major along j): src[i + NI*j] -> dst[j + NJ*i].
+ * This is the unpack kernel's shape with batch=1. */
+
+template <int TILE, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void t_scalar(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TILE][TILE + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t t_j = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TILE, j0 = t_j * TILE;
+    const int64_t ni = min((int64_t)TILE, NI - i0);
+    const int64_t nj = min((int64_t)TILE, NJ - j0);
+    for (int j = ty; j < nj; j += NROWS)
+        for (int i = tx; i < ni; i += 64)
+            tile[j][i] = src[(i0 + i) + NI * (j0 + j)];
+    __syncthreads();
+    for (int i = ty; i < ni; i += NROWS)
+        for (int j = tx; j < nj; j += 64)
+            dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
+}
+
+/* vectorized 16-B loads/stores: interior tiles only (caller guarantees
+ * NI,NJ multiples of TILE).  Lanes 0..31 load row 2*ty, lanes 32..63 row
+ * 2*ty+1 (uint4 = 2 f64 along i).  Write phase symmetric along j. */
+template <int TILE, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void t_vec(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TILE][TILE + 2];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int lane = tx & 31;      /* 0..31 */
+    const int half = tx >> 5;      /* 0 or 1 */
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t t_j = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TILE, j0 = t_j * TILE;
+
+    /* load: rows 2*r+half for r = ty, ty+NROWS, ... ; each lane one uint4 */
+    for (int r = ty; r < TILE / 2; r += NROWS) {
+        const int j = 2 * r + half;
+        const uint4 v = *(const uint4 *)&src[(i0 + 2 * lane) + NI * (j0 + j)];
+        tile[j][2 * lane] = ((const uint64_t *)&v)[0];
+        tile[j][2 * lane + 1] = ((const uint64_t *)&v)[1];
+    }
+    __syncthreads();
+    /* store: output rows i = 2*r+half; lane covers j = 2*lane, 2*lane+1 */
+    for (int r = ty; r < TILE / 2; r += NROWS) {
+        const int i = 2 * r + half;
+        uint4 v;
+        ((uint64_t *)&v)[0] = tile[2 * lane][i];
+        ((uint64_t *)&v)[1] = tile[2 * lane + 1][i];
+        *(uint4 *)&dst[(j0 + 2 * lane) + NJ * (i0 + i)] = v;
+    }
+}
+
+/* ---------------- harness ---------------- */
+
+static double bench(void (*launch)(void *, void *, int64_t, int), void *s,
+                    void *d, int64_t n, int arg, int reps, double bytes)
+{
+    hipEvent_t a, b;
+    CHK(hipEventCreate(&a));
+    CHK(hipEventCreate(&b));
+    launch(s, d, n, arg); /* warmup */
+    CHK(hipDeviceSynchronize());
+    double best = 1e30;
+    for (int r = 0; r < reps; r++) {
+        CHK(hipEventRecord(a));
+        launch(s, d, n, arg);
+        CHK(hipEventRecord(b));
+        CHK(hipEventSynchronize(b));
+        float ms;
+        CHK(hipEventElapsedTime(&ms, a, b));
+        if (ms < best) best = ms;
+    }
+    CHK(hipEventDestroy(a));
+    CHK(hipEventDestroy(b));
+    return bytes / (best * 1e-3) / 1e9;
+}
+
+#define NELEM_BYTES (4LL << 30) /* 4 GiB payload */
+
+int main()
+{
+    const int64_t bytes = NELEM_BYTES;
+    const int64_t n16 = bytes / 16;
+    void *s, *d;
+    CHK(hipMalloc(&s, bytes));
+    CHK(hipMalloc(&d, bytes));
+    CHK(hipMemset(s, 0x5A, bytes));
+    const double io = 2.0 * bytes;
+    const int reps = 5;
+
+    auto g = [](int64_t work, int perblk, int cap) {
+        int64_t b = (work + perblk - 1) / perblk;
+        if (cap && b > cap) b = cap;
+        return (int)b;
+    };
+
+#define RUN(name, launch_expr)                                               \
+    {                                                                        \
+        auto L = +[](void *ss, void *dd, int64_t n, int arg) {               \
+            (void)arg;                                                       \
+            launch_expr;                                                     \
+        };                                                                   \
+        printf("%-28s %8.1f GB/s\n", name, bench(L, s, d, n16, 0, reps, io)); \
+        fflush(stdout);                                                      \
+    }
+
+    RUN("1d gridstride cap2048", {
+        hipLaunchKernelGGL(c_gridstride, dim3(2048), dim3(256), 0, 0,
+                           (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d gridstride cap8192", {
+        hipLaunchKernelGGL(c_gridstride, dim3(8192), dim3(256), 0, 0,
+                           (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d direct full grid", {
+        hipLaunchKernelGGL(c_direct, dim3((uint32_t)((n + 255) / 256)),
+                           dim3(256), 0, 0, (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d direct 512thr", {
+        hipLaunchKernelGGL(c_direct512, dim3((uint32_t)((n + 511) / 512)),
+                           dim3(512), 0, 0, (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d nt cap2048", {
+        hipLaunchKernelGGL(c_nt, dim3(2048), dim3(256), 0, 0,
+                           (const v4u *)ss, (v4u *)dd, n);
+    });
+    RUN("1d unroll4 cap2048", {
+        hipLaunchKernelGGL(c_unroll4, dim3(2048), dim3(256), 0, 0,
+                           (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d unroll4 cap4096", {
+        hipLaunchKernelGGL(c_unroll4, dim3(4096), dim3(256), 0, 0,
+                           (const uint4 *)ss, (uint4 *)dd, n);
+    });
+    RUN("1d nt+unroll4 cap2048", {
+        hipLaunchKernelGGL(c_nt_unroll4, dim3(2048), dim3(256), 0, 0,
+                           (const v4u *)ss, (v4u *)dd, n);
+    });
+    RUN("1d nt+unroll4 cap4096", {
+        hipLaunchKernelGGL(c_nt_unroll4, dim3(4096), dim3(256), 0, 0,
+                           (const v4u *)ss, (v4u *)dd, n);
+    });
+
+    /* transpose probes: 16384 x 16384 f64 (2 GiB payload x2) */
+    {
+        const int64_t NI = 16384, NJ = 16384;
+        const double tio = 2.0 * NI * NJ * 8;
+        const int64_t nti64 = NI / 64;
+        const int64_t blocks64 = (NI / 64) * (NJ / 64);
+#define TRUN(name, kern, TI, NR)                                             \
+    {                                                                        \
+        const int64_t nti = NI / TI;                                         \
+        const int64_t blocks = (NI / TI) * (NJ / TI);                        \
+        hipEvent_t a, b;                                                     \
+        CHK(hipEventCreate(&a));                                             \
+        CHK(hipEventCreate(&b));                                             \
+        hipLaunchKernelGGL((kern<TI, NR>), dim3((uint32_t)blocks),           \
+                           dim3(64, NR), 0, 0, (const uint64_t *)s,          \
+                           (uint64_t *)d, NI, NJ, nti);                      \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < reps; r++) {                                     \
+            CHK(hipEventRecord(a));                                          \
+            hipLaunchKernelGGL((kern<TI, NR>), dim3((uint32_t)blocks),       \
+                               dim3(64, NR), 0, 0, (const uint64_t *)s,      \
+                               (uint64_t *)d, NI, NJ, nti);                  \
+            CHK(hipEventRecord(b));                                          \
+            CHK(hipEventSynchronize(b));                                     \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a, b));                             \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("%-28s %8.1f GB/s\n", name, tio / (best * 1e-3) / 1e9);       \
+        fflush(stdout);                                                      \
+        CHK(hipEventDestroy(a));                                             \
+        CHK(hipEventDestroy(b));                                             \
+    }
+        (void)nti64;
+        (void)blocks64;
+        TRUN("tr scalar 64x64 r8", t_scalar, 64, 8);
+        TRUN("tr scalar 64x64 r16", t_scalar, 64, 16);
+        TRUN("tr scalar 32x32 r8", t_scalar, 32, 8);
+        TRUN("tr scalar 128x128 r8", t_scalar, 128, 8);
+        TRUN("tr vec16 64x64 r8", t_vec, 64, 8);
+        TRUN("tr vec16 64x64 r4", t_vec, 64, 4);
+        TRUN("tr vec16 64x64 r16", t_vec, 64, 16);
+
+        /* correctness of t_vec 64x64 on a small pattern */
+        {
+            const int64_t ni = 256, nj = 192;
+            uint64_t *hs = (uint64_t *)malloc(ni * nj * 8);
+            uint64_t *hd = (uint64_t *)malloc(ni * nj * 8);
+            for (int64_t i = 0; i < ni * nj; i++) hs[i] = i * 2654435761ULL;
+            CHK(hipMemcpy(s, hs, ni * nj * 8, hipMemcpyHostToDevice));
+            CHK(hipMemset(d, 0xCC, ni * nj * 8));
+            hipLaunchKernelGGL((t_vec<64, 8>),
+                               dim3((uint32_t)((ni / 64) * (nj / 64))),
+                               dim3(64, 8), 0, 0, (const uint64_t *)s,
+                               (uint64_t *)d, ni, nj, ni / 64);
+            CHK(hipMemcpy(hd, d, ni * nj * 8, hipMemcpyDeviceToHost));
+            int64_t bad = 0;
+            for (int64_t j = 0; j < nj; j++)
+                for (int64_t i = 0; i < ni; i++)
+                    if (hd[j + nj * i] != hs[i + ni * j]) bad++;
+            printf("t_vec correctness: %s (%lld bad)\n",
+                   bad ? "FAIL" : "OK", (long long)bad);
+        }
+    }
+
+    CHK(hipFree(s));
+    CHK(hipFree(d));
+    return 0;
+}
