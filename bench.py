@@ -47,6 +47,9 @@ def parse_args():
     ap.add_argument("--permuted", action="store_true",
                     help="config-5 variant: output pencil memory-permuted "
                          "(1,2,0) — the PencilFFTs layout")
+    ap.add_argument("--double", action="store_true",
+                    help="config-4 variant: x->y->z chained double "
+                         "transpose (PencilFFTs pattern), deferred waits")
     ap.add_argument("--dtype", default="float64",
                     choices=["float64", "complex64"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -109,8 +112,9 @@ def main():
 
     topo = Topology(grid)
     Pi = Pencil(topo, dims, (1, 2))
-    po_perm = (1, 2, 0) if args.permuted else None
+    po_perm = (1, 2, 0) if (args.permuted or args.double) else None
     Po = Pencil(topo, dims, (0, 2), permute=po_perm)
+    Pz = Pencil(topo, dims, (0, 1), permute=(2, 1, 0)) if args.double else None
 
     gen = torch.Generator(device="cpu").manual_seed(0xC0FFEE + rank)
     n_in = Pi.length_local(rank)
@@ -124,12 +128,25 @@ def main():
 
     t = Transposition(dst, src)
     t.execute()  # builds native plan, allocates staging, inits RCCL comms
+    t2 = None
+    if args.double:
+        dstz = PencilArray(
+            Pz, rank,
+            torch.empty(Pz.length_local(rank), dtype=tdt, device=device))
+        t2 = Transposition(dstz, dst)
+        t2.execute()
     torch.cuda.synchronize()
 
     stream = torch.cuda.current_stream()
 
     def step(sync=False):
-        t._native.execute(src.data, dst.data, sync=sync)
+        # deferred waits (waitall=false pattern): both hops enqueue, the
+        # stream orders them
+        t._native.execute(src.data, dst.data, sync=False)
+        if t2 is not None:
+            t2._native.execute(dst.data, dstz.data, sync=False)
+        if sync:
+            torch.cuda.synchronize()
 
     for _ in range(args.warmup):
         step()
@@ -162,7 +179,8 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1e3
     global_bytes = math.prod(dims) * esz
-    gib_s = global_bytes / (elapsed / args.steps) / 2**30
+    nhops = 2 if args.double else 1
+    gib_s = nhops * global_bytes / (elapsed / args.steps) / 2**30
 
     if rank != 0:
         return
@@ -173,16 +191,25 @@ def main():
     # N>1 the dominant resource is the xGMI link: bytes crossing one link
     # per step = remote fraction of the local block, both directions.
     P_sub = grid[0]
+    hops = nhops
     if n_gpus == 1:
-        kernel_ms = gpu_ms / args.steps
-        algo_bytes = 2 * global_bytes  # per launch == per step at N=1
+        kernel_ms = gpu_ms / args.steps / hops
+        algo_bytes = 2 * global_bytes  # per launch == per hop at N=1
+        # traffic: measured with rocprofv3 PMC on this exact kernel+workload
+        # (profiles/r01_pmc_traffic.md): FETCH_SIZE*2 (gfx950 wide-coalesced
+        # correction) + WRITE_SIZE == algorithmic bytes, no re-reads.
+        measured_traffic = (algo_bytes
+                            if tuple(dims) == (1024, 1024, 1024) else None)
         roofline = {
             "bound": "hbm",
             "achieved": round(algo_bytes / (kernel_ms * 1e-3) / 1e9, 1),
             "peak": 8000.0,
             "unit": "GB/s",
             "frac": round(algo_bytes / (kernel_ms * 1e-3) / 8e12, 4),
-            "traffic": None,
+            "traffic": measured_traffic,
+            "traffic_source": ("rocprofv3 --pmc FETCH_SIZE/WRITE_SIZE, "
+                               "profiles/r01_pmc_traffic.md"
+                               if measured_traffic else None),
             "kernel": "fused local permuted copy (k_copy_1d / k_transpose_tile)",
         }
     else:
@@ -218,9 +245,12 @@ def main():
         "dtype": "f64" if args.dtype == "float64" else "c64",
         "data": "synthetic",
         "config": {
-            "workload": f"{sz} {args.dtype} x->y pencil transpose"
-                        + (" (memory-permuted output, PencilFFTs layout)"
-                           if args.permuted else ""),
+            "workload": f"{sz} {args.dtype} "
+                        + ("x->y->z double transpose (PencilFFTs pattern, "
+                           "deferred waits)" if args.double else
+                           "x->y pencil transpose"
+                           + (" (memory-permuted output, PencilFFTs layout)"
+                              if args.permuted else "")),
             "grid": f"{grid[0]}x{grid[1]}",
             "exchange": "rccl" if n_gpus > 1 else "none (local path)",
         },

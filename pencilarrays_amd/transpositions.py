@@ -109,21 +109,33 @@ class Transposition:
     # GPU execution — native HIP engine only.
     # ------------------------------------------------------------------
 
-    def _execute_torch_cuda(self):
+    def _execute_torch_cuda(self, sync: bool):
         from . import native
         if self._native is None:
             self._native = native.NativeTransposition(self)
-        self._native.execute(self.src.data, self.dest.data)
+        self._native.execute(self.src.data, self.dest.data, sync=sync)
 
     # ------------------------------------------------------------------
 
-    def execute(self):
+    def execute(self, sync: bool = True):
+        """Run the transpose.  ``sync=False`` is the reference's
+        ``transpose!(t; waitall=false)`` contract (Transpositions.jl:142-158):
+        the call returns with work enqueued on the stream; call :meth:`wait`
+        (== ``MPI.Waitall(t)``, :128-131) before reusing the source or
+        relying on the destination outside the stream."""
         if self.src.is_torch and self.src.data.is_cuda:
-            self._execute_torch_cuda()
+            self._execute_torch_cuda(sync)
         else:
             import torch.distributed as dist
             self._execute_numpy(use_dist=dist.is_available() and dist.is_initialized())
         return self.dest
+
+    def wait(self):
+        """MPI.Waitall(t) (Transpositions.jl:128-131)."""
+        if self._native is not None:
+            import torch
+            self._native.native.wait(torch.cuda.current_stream().cuda_stream)
+        return self
 
 
 def transpose_into(dest: PencilArray, src: PencilArray) -> PencilArray:
