@@ -25,10 +25,14 @@ from .plan import CopyDesc, TransposePlan, build_plan, normalize_desc
 from .transpositions import Transposition, run_transpose_sim, transpose_into
 from .gather import gather_dist, gather_sim
 
+# name parity with the reference's `gather` (gather.jl:17): the distributed
+# form, returning the global array on root and None elsewhere.
+gather = gather_dist
+
 __all__ = [
     "Topology", "dims_create", "Pencil", "PencilArray",
     "Transposition", "transpose_into", "run_transpose_sim",
-    "gather_sim", "gather_dist",
+    "gather", "gather_sim", "gather_dist",
     "CopyDesc", "TransposePlan", "build_plan", "normalize_desc",
     "identity_perm", "perm_apply", "perm_unapply", "perm_inv",
     "perm_relative", "perm_append",
