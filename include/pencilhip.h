@@ -80,10 +80,15 @@ void pa_comm_destroy(pa_comm *c);
  * pure data movement; dtype enters via size only).  extra_dims (E slowest
  * axes, arrays.jl:105-106) may be NULL with e = 0.
  * Unlike the reference, which re-plans on every transpose! call (:165-167),
- * a plan is reusable and allocation-free in steady state. */
+ * a plan is reusable and allocation-free in steady state.
+ * flags bit 0 (PA_PLAN_ALIASED): src and dst parents may alias (the
+ * reference's in-place / ManyPencilArray transposes, multiarrays.jl:106-143,
+ * Transpositions.jl:250-264): the self block then stages through the
+ * recv-buffer tail so every src read completes before any dst write. */
+#define PA_PLAN_ALIASED 1
 pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                          int64_t elem_size, int e, const int64_t *extra_dims,
-                         int rank, pa_plan **out);
+                         int rank, int flags, pa_plan **out);
 void pa_plan_destroy(pa_plan *p);
 
 /* Subgroup communicator for the exchange (required when the subgroup size
@@ -112,7 +117,8 @@ int pa_plan_nproc_sub(const pa_plan *p);      /* P (1 => purely local)  */
 int pa_plan_r_dim(const pa_plan *p);          /* R, or -1 if same decomp */
 int pa_plan_my_k(const pa_plan *p);
 pa_status pa_plan_block_info(const pa_plan *p, int k, int64_t out[8]);
-/* which: 0 = local fused copy, 1 = pack of peer k, 2 = unpack of peer k.
+/* which: 0 = local fused copy, 1 = pack of peer k, 2 = unpack of peer k,
+ * 3 = staged self pack (aliased mode), 4 = staged self unpack.
  * Returns the normalized strided-copy descriptor (element units):
  * nd, dims[nd], sstrides[nd], soffset, dstrides[nd], doffset. */
 pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,

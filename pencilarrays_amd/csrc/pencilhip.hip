@@ -533,6 +533,11 @@ struct pa_plan {
     std::vector<PeerBlockC> peers;
     bool has_local = false;
     CopyDescH local;
+    /* aliased (in-place) mode: self block staged through the recv tail
+     * (Transpositions.jl:250-264, :394-404) */
+    bool aliased = false;
+    bool has_self = false;
+    CopyDescH self_pack, self_unpack;
     int64_t send_total = 0, recv_total = 0; /* elements (remote blocks) */
     void *send_buf = nullptr, *recv_buf = nullptr;
     bool own_bufs = false;
@@ -626,6 +631,37 @@ static CopyDescH unpack_desc(const pa_plan &pl, const Range *grange,
     for (int e = 0; e < E; e++) dstr[n + e] = pst_o[n + e];
 
     return normalize_desc(n + E, bufdims, bufstr, bufoff, dstr, doff);
+}
+
+
+/* staged self block (aliased mode): pack src window -> recv tail, unpack
+ * recv tail -> dst window */
+static void staged_self_descs(pa_plan *pl, const Range *sr, const Range *rr,
+                              int64_t recv_off)
+{
+    const pa_pencil &Pi = pl->Pi;
+    const int n = Pi.n, e = pl->E;
+    int64_t dims[2 * MAXND], str[2 * MAXND], off;
+    int kk;
+    window_desc(Pi, pl->rank, pl->extra, sr, dims, str, &off, &kk);
+    int64_t cstr[2 * MAXND], acc = 1;
+    for (int i = 0; i < kk; i++) {
+        cstr[i] = acc;
+        acc *= dims[i];
+    }
+    pl->self_pack = normalize_desc(kk, dims, str, off, cstr, recv_off);
+    int64_t bdims[2 * MAXND];
+    for (int i = 0; i < n; i++)
+        bdims[i] = rr[Pi.perm[i]].hi - rr[Pi.perm[i]].lo;
+    for (int e2 = 0; e2 < e; e2++) bdims[n + e2] = pl->extra[e2];
+    int64_t bstr[2 * MAXND];
+    acc = 1;
+    for (int i = 0; i < n + e; i++) {
+        bstr[i] = acc;
+        acc *= bdims[i];
+    }
+    pl->self_unpack = unpack_desc(*pl, rr, bdims, bstr, recv_off);
+    pl->has_self = true;
 }
 
 extern "C" {
@@ -768,7 +804,7 @@ void pa_comm_destroy(pa_comm *c)
 
 pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                          int64_t elem_size, int e, const int64_t *extra_dims,
-                         int rank, pa_plan **out)
+                         int rank, int flags, pa_plan **out)
 {
     /* assert_compatible (Transpositions.jl:182-199) */
     if (pin->topo.m != pout->topo.m || pin->topo.dims != pout->topo.dims)
@@ -797,6 +833,7 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
     if (e) pl->extra.assign(extra_dims, extra_dims + e);
     pl->esz = elem_size;
     pl->R = R;
+    pl->aliased = (flags & 1) != 0;
 
     const int n = pin->n;
     const int64_t pex = prod_extra(*pl);
@@ -804,16 +841,22 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
     if (R < 0) {
         pl->P = 1;
         pl->myk = 0;
-        /* local path (transpose_impl!(::Nothing), Transpositions.jl:214-271):
-         * one fused copy/permute */
+        /* local path (transpose_impl!(::Nothing), Transpositions.jl:214-271;
+         * aliased variant stages through recv_buf, :250-264) */
         Range region[MAXND];
         axes_for_rank(*pin, rank, region);
         if (region_nelem(n, region) > 0) {
-            int64_t dims[2 * MAXND], str[2 * MAXND], off;
-            int k;
-            window_desc(*pin, rank, pl->extra, region, dims, str, &off, &k);
-            pl->local = unpack_desc(*pl, region, dims, str, off);
-            pl->has_local = true;
+            if (pl->aliased) {
+                staged_self_descs(pl, region, region, 0);
+                pl->recv_total = region_nelem(n, region) * pex;
+            } else {
+                int64_t dims[2 * MAXND], str[2 * MAXND], off;
+                int k;
+                window_desc(*pin, rank, pl->extra, region, dims, str, &off,
+                            &k);
+                pl->local = unpack_desc(*pl, region, dims, str, off);
+                pl->has_local = true;
+            }
         }
         *out = pl;
         return 0;
@@ -852,13 +895,22 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
             blk.recv_off = 0;
             /* fused self path: direct window->window permuted copy
              * (replaces :394-404 + :588-606; same values, half the HBM
-             * traffic) */
+             * traffic).  Aliased (in-place) mode stages through the recv
+             * tail like the reference so every src read precedes any dst
+             * write. */
             if (blk.send_n > 0) {
-                int64_t dims[2 * MAXND], str[2 * MAXND], off;
-                int kk;
-                window_desc(*pin, rank, pl->extra, sr, dims, str, &off, &kk);
-                pl->local = unpack_desc(*pl, rr, dims, str, off);
-                pl->has_local = true;
+                if (pl->aliased) {
+                    /* recv_off = length of remote recvs; filled below once
+                     * known: record block extents now, patch offset later */
+                    blk.recv_off = -1; /* placeholder, fixed after loop */
+                } else {
+                    int64_t dims[2 * MAXND], str[2 * MAXND], off;
+                    int kk;
+                    window_desc(*pin, rank, pl->extra, sr, dims, str, &off,
+                                &kk);
+                    pl->local = unpack_desc(*pl, rr, dims, str, off);
+                    pl->has_local = true;
+                }
             }
         } else {
             blk.send_off = isend;
@@ -897,6 +949,23 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
     }
     pl->send_total = isend;
     pl->recv_total = irecv;
+    if (pl->aliased) {
+        /* self block at the END of the recv buffer (:394-404) */
+        PeerBlockC &self = pl->peers[pl->myk];
+        self.recv_off = irecv;
+        if (self.send_n > 0) {
+            Range pc_sr[MAXND], pc_rr[MAXND];
+            Range axp_o[MAXND], axp_i[MAXND];
+            axes_for_coords(*pout, coords, axp_o);
+            axes_for_coords(*pin, coords, axp_i);
+            for (int d = 0; d < n; d++) {
+                range_intersect(axl_i[d], axp_o[d], &pc_sr[d]);
+                range_intersect(axl_o[d], axp_i[d], &pc_rr[d]);
+            }
+            staged_self_descs(pl, pc_sr, pc_rr, irecv);
+            pl->recv_total = irecv + self.send_n;
+        }
+    }
     *out = pl;
     return 0;
 }
@@ -961,13 +1030,20 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
         p->own_bufs = true;
     }
 
-    /* 1. pack every remote block (Transpositions.jl:346-431) */
+    /* 1. pack every remote block (Transpositions.jl:346-431); in aliased
+     * mode also stage the self block into the recv tail before any dst
+     * write (:394-404) */
     for (auto &blk : p->peers)
         if (blk.has_pack) {
             pa_status st =
                 launch_desc(blk.pack, p->esz, src_parent, p->send_buf, stream);
             if (st) return st;
         }
+    if (p->has_self) {
+        pa_status st = launch_desc(p->self_pack, p->esz, src_parent,
+                                   p->recv_buf, stream);
+        if (st) return st;
+    }
 
     /* 2. exchange: grouped ncclSend/ncclRecv over xGMI on a dedicated comm
      * stream ordered after the pack kernels (replaces :419-428/:463-479;
@@ -1013,10 +1089,15 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
     }
 
     /* 3. fused local/self copy — on the caller's stream, overlapping the
-     * exchange */
+     * exchange (aliased mode: unpack the staged self block instead) */
     if (p->has_local) {
         pa_status st =
             launch_desc(p->local, p->esz, src_parent, dst_parent, stream);
+        if (st) return st;
+    }
+    if (p->has_self) {
+        pa_status st = launch_desc(p->self_unpack, p->esz, p->recv_buf,
+                                   dst_parent, stream);
         if (st) return st;
     }
 
@@ -1070,6 +1151,12 @@ pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,
     if (which == 0) {
         if (!p->has_local) return fail("no local copy in plan");
         d = &p->local;
+    } else if (which == 3) {
+        if (!p->has_self) return fail("no staged self pack in plan");
+        d = &p->self_pack;
+    } else if (which == 4) {
+        if (!p->has_self) return fail("no staged self unpack in plan");
+        d = &p->self_unpack;
     } else {
         if (p->R < 0 || k < 0 || k >= (int)p->peers.size())
             return fail("no peer block %d", k);

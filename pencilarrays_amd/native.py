@@ -58,7 +58,7 @@ class NativePlan:
     building is host-only (works without a GPU); execute needs one."""
 
     def __init__(self, Pi, Po, rank: int, elem_size: int,
-                 extra_dims: Tuple[int, ...] = ()):
+                 extra_dims: Tuple[int, ...] = (), aliased: bool = False):
         lib = self.lib = load()
         topo = Pi.topology
         m = topo.ndims
@@ -84,6 +84,7 @@ class NativePlan:
         _check(lib, lib.pa_plan_create(
             self._pi, self._po, I64(elem_size), e,
             (I64 * e)(*extra_dims) if e else None, rank,
+            1 if aliased else 0,
             ctypes.byref(self._plan)), "pa_plan_create")
         self.nproc_sub = lib.pa_plan_nproc_sub(self._plan)
         self.r_dim = lib.pa_plan_r_dim(self._plan)
@@ -212,7 +213,7 @@ class NativeTransposition:
         esz = src.data.element_size()
         self.np_plan = plan
         self.native = NativePlan(plan.Pi, plan.Po, plan.rank, esz,
-                                 plan.extra_dims)
+                                 plan.extra_dims, aliased=plan.aliased)
         sb, rb = self.native.buffer_sizes()
         dev = src.data.device
         self._send = torch.empty(max(sb, 1), dtype=torch.uint8, device=dev)

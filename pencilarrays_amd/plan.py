@@ -115,7 +115,15 @@ class TransposePlan:
     peers: List[PeerBlock] = field(default_factory=list)
     local: Optional[CopyDesc] = None  # fused self/local permuted copy
     send_nelem_total: int = 0
-    recv_nelem_total: int = 0       # staging for remote blocks only
+    recv_nelem_total: int = 0       # staging incl. self tail when aliased
+    aliased: bool = False
+    # aliased (in-place) mode: the self block stages through the recv-buffer
+    # tail instead of the fused direct copy (the reference's aliasing path:
+    # Transpositions.jl:250-264 for the local case, :394-404 placement for
+    # the distributed case), so every read of src completes before any write
+    # of dst.
+    self_pack: Optional[CopyDesc] = None    # src -> recv_buf tail
+    self_unpack: Optional[CopyDesc] = None  # recv_buf tail -> dst
 
     @property
     def subgroup_global_ranks(self) -> List[int]:
@@ -147,7 +155,8 @@ def _window_desc_src(p: Pencil, rank: int, region: Region,
 
 
 def build_plan(Pi: Pencil, Po: Pencil, rank: int,
-               extra_dims: Tuple[int, ...] = ()) -> TransposePlan:
+               extra_dims: Tuple[int, ...] = (),
+               aliased: bool = False) -> TransposePlan:
     extra_dims = tuple(int(e) for e in extra_dims)
     R = Pi.transpose_dim(Po)  # validates compatibility
     n = Pi.ndims
@@ -165,7 +174,7 @@ def build_plan(Pi: Pencil, Po: Pencil, rank: int,
     plan = TransposePlan(
         rank=rank, Pi=Pi, Po=Po, extra_dims=extra_dims,
         r_dim=R, nproc_sub=1 if R is None else Pi.topology.dims[R],
-        my_k=0,
+        my_k=0, aliased=aliased,
     )
 
     mem_o, pst_o = _parent_strides(Po, rank, extra_dims)
@@ -186,13 +195,34 @@ def build_plan(Pi: Pencil, Po: Pencil, rank: int,
             dstrides=dstrides, doffset=doffset,
         ))
 
+    def staged_self(region_in: Region, region_out: Region, recv_off: int):
+        """Self block via the recv-buffer tail (in-place/aliased mode)."""
+        sdims, sst, soff = _window_desc_src(Pi, rank, region_in, extra_dims)
+        plan.self_pack = normalize_desc(CopyDesc(
+            dims=sdims, sstrides=sst, soffset=soff,
+            dstrides=_colmajor_strides(sdims), doffset=recv_off,
+        ))
+        bdims = (perm_apply(Pi.perm, region_lengths(region_out))
+                 + tuple(extra_dims))
+        dst_local = Po.to_local(rank, region_out, memory_order=True)
+        plan.self_unpack = unpack_like(
+            bdims, dst_local, _colmajor_strides(bdims), recv_off)
+
     if R is None:
         # Same decomposition: plain copy or local permutation
-        # (transpose_impl!(::Nothing), Transpositions.jl:214-271).
+        # (transpose_impl!(::Nothing), Transpositions.jl:214-271; in-place
+        # variant stages through recv_buf, :250-264).
         region = Pi.axes_for_rank(rank)
-        sdims, sst, soff = _window_desc_src(Pi, rank, region, extra_dims)
-        dst_local = Po.to_local(rank, region, memory_order=True)
-        plan.local = unpack_like(sdims, dst_local, sst, soff)
+        if region_nelem(region) > 0:
+            if aliased:
+                staged_self(region, region, 0)
+                plan.recv_nelem_total = (region_nelem(region)
+                                         * math.prod(extra_dims or (1,)))
+            else:
+                sdims, sst, soff = _window_desc_src(Pi, rank, region,
+                                                    extra_dims)
+                dst_local = Po.to_local(rank, region, memory_order=True)
+                plan.local = unpack_like(sdims, dst_local, sst, soff)
         return plan
 
     topo = Pi.topology
@@ -233,12 +263,19 @@ def build_plan(Pi: Pencil, Po: Pencil, rank: int,
             # Fused self path: direct src-window -> dst-window permuted copy
             # (replaces the reference's copy to the recv_buf tail :394-404
             # followed by copy_permuted! — same values, half the HBM traffic).
+            # In aliased (in-place) mode the direct copy would read windows
+            # the unpacks overwrite, so the self block stages through the
+            # recv tail exactly like the reference.
             assert nr == length_self and ns == length_self
             blk.recv_offset = length_recv_remote  # kept for reference parity
             if ns > 0:
-                sdims, sst, soff = _window_desc_src(Pi, rank, srange, extra_dims)
-                dst_local = Po.to_local(rank, rrange, memory_order=True)
-                plan.local = unpack_like(sdims, dst_local, sst, soff)
+                if aliased:
+                    staged_self(srange, rrange, length_recv_remote)
+                else:
+                    sdims, sst, soff = _window_desc_src(Pi, rank, srange,
+                                                        extra_dims)
+                    dst_local = Po.to_local(rank, rrange, memory_order=True)
+                    plan.local = unpack_like(sdims, dst_local, sst, soff)
         else:
             blk.send_offset = isend
             blk.recv_offset = irecv
@@ -261,6 +298,6 @@ def build_plan(Pi: Pencil, Po: Pencil, rank: int,
         plan.peers.append(blk)
 
     plan.send_nelem_total = isend
-    plan.recv_nelem_total = irecv
     assert irecv == length_recv_remote
+    plan.recv_nelem_total = irecv + (length_self if aliased else 0)
     return plan

@@ -34,6 +34,25 @@ from .plan import TransposePlan, build_plan
 MPI_TAG = 42  # Transpositions.jl:469
 
 
+def _arrays_alias(a, b) -> bool:
+    """Base.mightalias equivalent (Transpositions.jl:250): do the two parent
+    buffers share memory?  (ManyPencilArray in-place transposes.)"""
+    if isinstance(a, np.ndarray) and isinstance(b, np.ndarray):
+        return np.shares_memory(a, b)
+    if not isinstance(a, np.ndarray) and not isinstance(b, np.ndarray):
+        try:
+            sa, sb = a.untyped_storage(), b.untyped_storage()
+            if sa.data_ptr() != sb.data_ptr():
+                # conservative overlap check on address ranges
+                a0, a1 = a.data_ptr(), a.data_ptr() + a.nbytes
+                b0, b1 = b.data_ptr(), b.data_ptr() + b.nbytes
+                return a0 < b1 and b0 < a1
+            return True
+        except Exception:
+            return False
+    return False
+
+
 class Transposition:
     def __init__(self, dest: PencilArray, src: PencilArray):
         if dest.extra_dims != src.extra_dims:
@@ -44,8 +63,10 @@ class Transposition:
             raise ValueError("dest and src must live on the same rank")
         self.src = src
         self.dest = dest
+        self.aliased = _arrays_alias(src.data, dest.data)
         self.plan: TransposePlan = build_plan(
-            src.pencil, dest.pencil, src.rank, src.extra_dims)
+            src.pencil, dest.pencil, src.rank, src.extra_dims,
+            aliased=self.aliased)
         self._native = None  # set lazily for the GPU path
 
     # ------------------------------------------------------------------
@@ -61,6 +82,11 @@ class Transposition:
         if plan.r_dim is None or plan.nproc_sub == 1:
             if plan.local is not None:
                 apply_copy(plan.local, src_flat, dst_flat)
+            elif plan.self_pack is not None:
+                # in-place: stage through recv buffer (Transpositions.jl:250-264)
+                recv_buf = np.empty(plan.recv_nelem_total, dtype=src_flat.dtype)
+                apply_copy(plan.self_pack, src_flat, recv_buf)
+                apply_copy(plan.self_unpack, recv_buf, dst_flat)
             return
 
         if not use_dist:
@@ -74,10 +100,14 @@ class Transposition:
         send_buf = np.empty(plan.send_nelem_total, dtype=src_flat.dtype)
         recv_buf = np.empty(plan.recv_nelem_total, dtype=src_flat.dtype)
 
-        # 1. pack all remote blocks (Transpositions.jl:346-431)
+        # 1. pack all remote blocks (Transpositions.jl:346-431); in aliased
+        # mode also stage the self block into the recv tail BEFORE any write
+        # of dst (:394-404)
         for blk in plan.peers:
             if blk.pack is not None:
                 apply_copy(blk.pack, src_flat, send_buf)
+        if plan.self_pack is not None:
+            apply_copy(plan.self_pack, src_flat, recv_buf)
 
         # 2. exchange: per-peer nonblocking send/recv (:463-479)
         reqs = []
@@ -96,6 +126,8 @@ class Transposition:
         # 3. fused local (self) block overlaps the exchange (:394-404 + :530)
         if plan.local is not None:
             apply_copy(plan.local, src_flat, dst_flat)
+        elif plan.self_unpack is not None:
+            apply_copy(plan.self_unpack, recv_buf, dst_flat)
 
         for r in reqs:
             r.wait()
@@ -154,7 +186,8 @@ def transpose_into(dest: PencilArray, src: PencilArray) -> PencilArray:
 def run_transpose_sim(dests: Sequence[PencilArray],
                       srcs: Sequence[PencilArray]) -> None:
     nranks = len(srcs)
-    plans = [build_plan(s.pencil, d.pencil, r, s.extra_dims)
+    plans = [build_plan(s.pencil, d.pencil, r, s.extra_dims,
+                        aliased=_arrays_alias(s.data, d.data))
              for r, (d, s) in enumerate(zip(dests, srcs))]
 
     send_bufs = [np.empty(p.send_nelem_total, dtype=srcs[r].data.dtype)
@@ -162,13 +195,18 @@ def run_transpose_sim(dests: Sequence[PencilArray],
     recv_bufs = [np.empty(p.recv_nelem_total, dtype=srcs[r].data.dtype)
                  for r, p in enumerate(plans)]
 
-    # pack on every rank
+    # pack on every rank (incl. staged self blocks, before any dst write)
     for r, p in enumerate(plans):
         for blk in p.peers:
             if blk.pack is not None:
                 apply_copy(blk.pack, srcs[r].data, send_bufs[r])
+        if p.self_pack is not None:
+            apply_copy(p.self_pack, srcs[r].data, recv_bufs[r])
+    for r, p in enumerate(plans):
         if p.local is not None:
             apply_copy(p.local, srcs[r].data, dests[r].data)
+        elif p.self_unpack is not None:
+            apply_copy(p.self_unpack, recv_bufs[r], dests[r].data)
 
     # exchange: copy each send block into the receiver's recv buffer
     for r, p in enumerate(plans):

@@ -210,6 +210,36 @@ def test_multirank_sim_on_one_gpu(cfg):
         assert np.array_equal(dsts[r].cpu().numpy(), exp[r]), f"rank {r}"
 
 
+def test_inplace_world1_on_gpu():
+    """Aliased (in-place) transpose through pa_transpose_execute: staged
+    self path on device, bit-exact vs oracle (Transpositions.jl:250-264)."""
+    dims = (64, 48, 40)
+    topo = Topology((1, 1))
+    p1 = Pencil(topo, dims, (1, 2))
+    p2 = Pencil(topo, dims, (0, 2), permute=(1, 2, 0))
+    n = max(p1.length_local(0), p2.length_local(0))
+    buf = torch.empty(n, dtype=torch.float64, device="cuda:0")
+    rng = np.random.default_rng(11)
+    src_np = rng.standard_normal(p1.length_local(0))
+    buf[:p1.length_local(0)] = torch.from_numpy(src_np).cuda()
+    src = PencilArray(p1, 0, buf[:p1.length_local(0)])
+    dst = PencilArray(p2, 0, buf[:p2.length_local(0)])
+    t = Transposition(dst, src)
+    assert t.aliased
+    t.execute()
+    torch.cuda.synchronize()
+    exp = orc.transpose_oracle([src_np], dims, (1, 1), (1, 2), (0, 1, 2),
+                               (0, 2), (1, 2, 0), ())[0]
+    assert np.array_equal(dst.data.cpu().numpy(), exp)
+    # and back, in place again
+    back_exp = src_np
+    t2 = Transposition(src, dst)
+    assert t2.aliased
+    t2.execute()
+    torch.cuda.synchronize()
+    assert np.array_equal(src.data.cpu().numpy(), back_exp)
+
+
 def test_rccl_single_rank_bootstrap():
     lib = native.load()
     import ctypes

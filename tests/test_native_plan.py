@@ -60,6 +60,28 @@ def test_native_plan_matches_python(cfg):
                 assert _desc_tuple(blk.unpack) == nat.copydesc(2, blk.peer_k)
 
 
+@pytest.mark.parametrize("cfg", SWEEP[:8], ids=lambda c: f"{c[0]}x{c[1]}")
+def test_native_aliased_plan_matches_python(cfg):
+    dims, pdims, di, pi, do, po, extra, dtype = cfg
+    esz = np.dtype(dtype).itemsize
+    topo = Topology(pdims)
+    Pi = Pencil(topo, dims, di, permute=pi)
+    Po = Pencil(topo, dims, do, permute=po)
+    for rank in range(topo.nranks):
+        py = build_plan(Pi, Po, rank, extra, aliased=True)
+        nat = native.NativePlan(Pi, Po, rank, esz, extra, aliased=True)
+        sb, rb = nat.buffer_sizes()
+        assert sb == py.send_nelem_total * esz
+        assert rb == py.recv_nelem_total * esz
+        assert nat.copydesc(0) is None  # no fused local in aliased mode
+        assert _desc_tuple(py.self_pack) == nat.copydesc(3)
+        assert _desc_tuple(py.self_unpack) == nat.copydesc(4)
+        if py.r_dim is not None:
+            for blk in py.peers:
+                assert _desc_tuple(blk.pack) == nat.copydesc(1, blk.peer_k)
+                assert _desc_tuple(blk.unpack) == nat.copydesc(2, blk.peer_k)
+
+
 def test_plan_create_rejects_two_hop():
     topo = Topology((2, 2))
     Pi = Pencil(topo, (16, 21, 41), (1, 2))

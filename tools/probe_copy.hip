@@ -124,6 +124,50 @@ __global__ __launch_bounds__(64 * NROWS) void t_scalar(
             dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
 }
 
+/* rectangular tile: TI along src-contiguous axis, TJ along dst-contiguous
+ * axis (longer reads vs writes tradeoff). */
+template <int TI, int TJ, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void t_rect(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TJ][TI + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t t_j = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TI, j0 = t_j * TJ;
+    for (int j = ty; j < TJ; j += NROWS)
+        for (int i = tx; i < TI; i += 64)
+            tile[j][i] = src[(i0 + i) + NI * (j0 + j)];
+    __syncthreads();
+    for (int i = ty; i < TI; i += NROWS)
+        for (int j = tx; j < TJ; j += 64)
+            dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
+}
+
+/* 64x64 with nontemporal global accesses */
+typedef unsigned long v1u64 __attribute__((ext_vector_type(1)));
+template <int TILE, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void t_nt(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TILE][TILE + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t t_j = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TILE, j0 = t_j * TILE;
+    for (int j = ty; j < TILE; j += NROWS)
+        for (int i = tx; i < TILE; i += 64)
+            tile[j][i] =
+                __builtin_nontemporal_load(&src[(i0 + i) + NI * (j0 + j)]);
+    __syncthreads();
+    for (int i = ty; i < TILE; i += NROWS)
+        for (int j = tx; j < TILE; j += 64)
+            __builtin_nontemporal_store(tile[j][i],
+                                        &dst[(j0 + j) + NJ * (i0 + i)]);
+}
+
 /* vectorized 16-B loads/stores: interior tiles only (caller guarantees
  * NI,NJ multiples of TILE).  Lanes 0..31 load row 2*ty, lanes 32..63 row
  * 2*ty+1 (uint4 = 2 f64 along i).  Write phase symmetric along j. */
@@ -292,6 +336,43 @@ int main()
         TRUN("tr vec16 64x64 r8", t_vec, 64, 8);
         TRUN("tr vec16 64x64 r4", t_vec, 64, 4);
         TRUN("tr vec16 64x64 r16", t_vec, 64, 16);
+        TRUN("tr nt 64x64 r8", t_nt, 64, 8);
+
+#define TRUN2(name, TI, TJ, NR)                                              \
+    {                                                                        \
+        const int64_t nti = NI / TI;                                         \
+        const int64_t blocks = (NI / TI) * (NJ / TJ);                        \
+        hipEvent_t a, b;                                                     \
+        CHK(hipEventCreate(&a));                                             \
+        CHK(hipEventCreate(&b));                                             \
+        hipLaunchKernelGGL((t_rect<TI, TJ, NR>), dim3((uint32_t)blocks),     \
+                           dim3(64, NR), 0, 0, (const uint64_t *)s,          \
+                           (uint64_t *)d, NI, NJ, nti);                      \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < reps; r++) {                                     \
+            CHK(hipEventRecord(a));                                          \
+            hipLaunchKernelGGL((t_rect<TI, TJ, NR>), dim3((uint32_t)blocks), \
+                               dim3(64, NR), 0, 0, (const uint64_t *)s,      \
+                               (uint64_t *)d, NI, NJ, nti);                  \
+            CHK(hipEventRecord(b));                                          \
+            CHK(hipEventSynchronize(b));                                     \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a, b));                             \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("%-28s %8.1f GB/s\n", name, tio / (best * 1e-3) / 1e9);       \
+        fflush(stdout);                                                      \
+        CHK(hipEventDestroy(a));                                             \
+        CHK(hipEventDestroy(b));                                             \
+    }
+        TRUN2("tr rect 128x64 r8", 128, 64, 8);
+        TRUN2("tr rect 64x128 r8", 64, 128, 8);
+        TRUN2("tr rect 128x32 r8", 128, 32, 8);
+        TRUN2("tr rect 32x128 r8", 32, 128, 8);
+        TRUN2("tr rect 128x64 r16", 128, 64, 16);
+        TRUN2("tr rect 256x64 r8", 256, 64, 8);
+        TRUN2("tr rect 64x64 r8 (ctl)", 64, 64, 8);
 
         /* correctness of t_vec 64x64 on a small pattern */
         {
