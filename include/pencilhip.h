@@ -125,6 +125,14 @@ pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,
                            int64_t *dims, int64_t *sstr, int64_t *soff,
                            int64_t *dstr, int64_t *doff);
 
+/* ---- reductions (src/reductions.jl:9-38) ----------------------------- */
+/* One ncclAllReduce over a communicator (normally the FULL topology comm):
+ * the collective behind the reference's mapreduce/any/all.
+ * dtype: 0=f64, 1=f32, 2=i64, 3=i32, 4=u8; op: 0=sum, 1=prod, 2=min, 3=max.
+ * In-place allowed (send == recv). */
+pa_status pa_allreduce(pa_comm *c, const void *sendbuf, void *recvbuf,
+                       int64_t count, int dtype, int op, void *stream);
+
 /* ---- standalone device copy (used by tests/benchmarks) --------------- */
 /* Execute one strided-copy descriptor on device (same kernels the plan
  * uses).  All strides/offsets/dims in elements of elem_size bytes. */

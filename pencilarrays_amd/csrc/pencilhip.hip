@@ -152,19 +152,17 @@ __global__ __launch_bounds__(256) void k_copy_generic(const T *__restrict__ src,
  * is exactly this kernel; the reference's GPU path does a generic
  * permutedims! plus an extra temporary copy (:651-667 "TODO avoid
  * allocation") — here it is one kernel, no temporary. */
-template <typename T, int TILE, int NROWS>
+template <typename T, int TILE_I, int TILE_J, int NROWS>
 __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
     const T *__restrict__ src, T *__restrict__ dst, DescDev d, int ta,
     int64_t ntile_i, int64_t ntile_j)
 {
-    __shared__ T tile[TILE][TILE + 1];
+    __shared__ T tile[TILE_J][TILE_I + 1];
 
     const int tx = threadIdx.x; /* 0..63  : fast axis */
     const int ty = threadIdx.y; /* 0..NROWS-1 */
 
     const int64_t tiles_per_batch = ntile_i * ntile_j;
-    const int64_t nbatch_tiles = (int64_t)gridDim.x;
-    (void)nbatch_tiles;
 
     int64_t bid = blockIdx.x;
     const int64_t tij = bid % tiles_per_batch;
@@ -182,10 +180,10 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
         do_b += j * d.dstr[a];
     }
 
-    const int64_t i0 = t_i * TILE;       /* along axis 0  */
-    const int64_t j0 = t_j * TILE;       /* along axis ta */
-    const int64_t ni = d.dims[0] - i0 < TILE ? d.dims[0] - i0 : TILE;
-    const int64_t nj = d.dims[ta] - j0 < TILE ? d.dims[ta] - j0 : TILE;
+    const int64_t i0 = t_i * TILE_I;     /* along axis 0  */
+    const int64_t j0 = t_j * TILE_J;     /* along axis ta */
+    const int64_t ni = d.dims[0] - i0 < TILE_I ? d.dims[0] - i0 : TILE_I;
+    const int64_t nj = d.dims[ta] - j0 < TILE_J ? d.dims[ta] - j0 : TILE_J;
 
     /* load: lanes sweep axis 0 (src-contiguous), rows sweep axis ta */
     {
@@ -377,33 +375,38 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
 
     if (ta > 0 && (esz == 4 || esz == 8 || esz == 16)) {
         DescDev dd = to_dev(dn);
-        constexpr int TILE = 64, NROWS = 8;
-        const int64_t nti = (dn.dims[0] + TILE - 1) / TILE;
-        const int64_t ntj = (dn.dims[ta] + TILE - 1) / TILE;
         int64_t nbatch = 1;
         for (int a = 1; a < dn.nd; a++)
             if (a != ta) nbatch *= dn.dims[a];
-        const int64_t blocks = nti * ntj * nbatch;
-        if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
-        if (esz == 8)
-            hipLaunchKernelGGL((k_transpose_tile<uint64_t, TILE, NROWS>),
-                               dim3((uint32_t)blocks), dim3(64, NROWS), 0,
-                               stream, (const uint64_t *)s, (uint64_t *)d, dd,
-                               ta, nti, ntj);
-        else if (esz == 4)
-            hipLaunchKernelGGL((k_transpose_tile<uint32_t, TILE, NROWS>),
-                               dim3((uint32_t)blocks), dim3(64, NROWS), 0,
-                               stream, (const uint32_t *)s, (uint32_t *)d, dd,
-                               ta, nti, ntj);
-        else {
-            constexpr int TILE16 = 32; /* 16-B elements: 32×32 = 16 KiB+pad */
-            const int64_t nti2 = (dn.dims[0] + TILE16 - 1) / TILE16;
-            const int64_t ntj2 = (dn.dims[ta] + TILE16 - 1) / TILE16;
-            const int64_t blocks2 = nti2 * ntj2 * nbatch;
-            hipLaunchKernelGGL((k_transpose_tile<uint4, TILE16, NROWS>),
-                               dim3((uint32_t)blocks2), dim3(64, NROWS), 0,
+        /* tile shapes probe-measured on MI355X (profiles/r01_probe_copy2):
+         * 8-B elements: 128(i)×64(j) r16 = 5453 GB/s (vs 64×64 r8 5269);
+         * 16-B: 32×32 (LDS-bounded).  Long reads, 512-B write bursts. */
+        if (esz == 8 || esz == 4) {
+            constexpr int TI = 128, TJ = 64, NR = 16;
+            const int64_t nti = (dn.dims[0] + TI - 1) / TI;
+            const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
+            const int64_t blocks = nti * ntj * nbatch;
+            if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
+            if (esz == 8)
+                hipLaunchKernelGGL((k_transpose_tile<uint64_t, TI, TJ, NR>),
+                                   dim3((uint32_t)blocks), dim3(64, NR), 0,
+                                   stream, (const uint64_t *)s, (uint64_t *)d,
+                                   dd, ta, nti, ntj);
+            else
+                hipLaunchKernelGGL((k_transpose_tile<uint32_t, TI, TJ, NR>),
+                                   dim3((uint32_t)blocks), dim3(64, NR), 0,
+                                   stream, (const uint32_t *)s, (uint32_t *)d,
+                                   dd, ta, nti, ntj);
+        } else {
+            constexpr int TI = 32, TJ = 32, NR = 8;
+            const int64_t nti = (dn.dims[0] + TI - 1) / TI;
+            const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
+            const int64_t blocks = nti * ntj * nbatch;
+            if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
+            hipLaunchKernelGGL((k_transpose_tile<uint4, TI, TJ, NR>),
+                               dim3((uint32_t)blocks), dim3(64, NR), 0,
                                stream, (const uint4 *)s, (uint4 *)d, dd, ta,
-                               nti2, ntj2);
+                               nti, ntj);
         }
         HIP_CHECK(hipGetLastError());
         return 0;
@@ -1178,6 +1181,21 @@ pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,
     }
     *soff = d->soff;
     *doff = d->doff;
+    return 0;
+}
+
+/* ---- reductions ---------------------------------------------------- */
+
+pa_status pa_allreduce(pa_comm *c, const void *sendbuf, void *recvbuf,
+                       int64_t count, int dtype, int op, void *stream)
+{
+    static const ncclDataType_t dts[] = {ncclFloat64, ncclFloat32, ncclInt64,
+                                         ncclInt32, ncclUint8};
+    static const ncclRedOp_t ops[] = {ncclSum, ncclProd, ncclMin, ncclMax};
+    if (dtype < 0 || dtype > 4) return fail("bad dtype %d", dtype);
+    if (op < 0 || op > 3) return fail("bad op %d", op);
+    NCCL_CHECK(ncclAllReduce(sendbuf, recvbuf, (size_t)count, dts[dtype],
+                             ops[op], c->comm, (hipStream_t)stream));
     return 0;
 }
 
