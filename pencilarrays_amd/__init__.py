@@ -24,6 +24,8 @@ from .array import PencilArray
 from .plan import CopyDesc, TransposePlan, build_plan, normalize_desc
 from .transpositions import Transposition, run_transpose_sim, transpose_into
 from .multiarrays import ManyPencilArray
+from .pencilio import MPIIOFile
+from . import reductions
 from .gather import gather_dist, gather_sim
 
 # name parity with the reference's `gather` (gather.jl:17): the distributed
@@ -33,6 +35,7 @@ gather = gather_dist
 __all__ = [
     "Topology", "dims_create", "Pencil", "PencilArray",
     "Transposition", "transpose_into", "run_transpose_sim", "ManyPencilArray",
+    "MPIIOFile", "reductions",
     "gather", "gather_sim", "gather_dist",
     "CopyDesc", "TransposePlan", "build_plan", "normalize_desc",
     "identity_perm", "perm_apply", "perm_unapply", "perm_inv",
