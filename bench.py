@@ -177,6 +177,10 @@ def main():
         dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
         elapsed = float(tmax.item())
 
+    if dist:
+        dist.barrier()
+        dist.destroy_process_group()
+
     ms_per_step = elapsed / args.steps * 1e3
     global_bytes = math.prod(dims) * esz
     nhops = 2 if args.double else 1

@@ -100,5 +100,21 @@ class PencilArray:
         s = self.pencil.size_local(self.rank, memory_order)
         return tuple(s) + self.extra_dims
 
+    def similar(self, pencil: Optional[Pencil] = None, dtype=None) -> "PencilArray":
+        """similar(x, [p::Pencil]) (arrays.jl:287-303): a new uninitialised
+        PencilArray with the same (or the given) decomposition, same backend
+        and device."""
+        p = self.pencil if pencil is None else pencil
+        if self.is_torch:
+            import torch
+            dt = self.data.dtype if dtype is None else dtype
+            return PencilArray.empty(p, self.rank, dtype=dt,
+                                     extra_dims=self.extra_dims,
+                                     backend="torch",
+                                     device=self.data.device)
+        dt = self.data.dtype if dtype is None else dtype
+        return PencilArray.empty(p, self.rank, dtype=dt,
+                                 extra_dims=self.extra_dims)
+
     def __len__(self):
         return int(math.prod(self.mem_dims))
