@@ -240,6 +240,78 @@ def test_inplace_world1_on_gpu():
     assert np.array_equal(src.data.cpu().numpy(), back_exp)
 
 
+def test_randomized_descriptor_sweep():
+    """Property sweep: 15 seeded-random configurations (dims, grid, decomps,
+    perms, dtype) — every plan descriptor executed on device vs the CPU
+    executor, whole-buffer sentinel compare."""
+    import ctypes
+    import itertools
+    rng = np.random.default_rng(20260915)
+    lib = native.load()
+    I64 = ctypes.c_int64
+    from pencilarrays_amd.plan import build_plan as bp
+
+    def dev_copy(desc, esz, src_t, dst_t):
+        nd = len(desc.dims)
+        st = lib.pa_device_copy(
+            nd, (I64 * nd)(*desc.dims), (I64 * nd)(*desc.sstrides),
+            I64(desc.soffset), (I64 * nd)(*desc.dstrides), I64(desc.doffset),
+            I64(esz), ctypes.c_void_p(src_t.data_ptr()),
+            ctypes.c_void_p(dst_t.data_ptr()), None)
+        assert st == 0, lib.pa_last_error().decode()
+
+    for trial in range(15):
+        nd = int(rng.integers(2, 5))
+        dims = tuple(int(rng.integers(1, 24)) for _ in range(nd))
+        m = int(rng.integers(1, min(nd, 2) + 1))
+        pdims = tuple(int(rng.integers(1, 4)) for _ in range(m))
+        all_dims = list(range(nd))
+        di = tuple(rng.permutation(all_dims)[:m].tolist())
+        do = list(di)
+        avail = [d for d in all_dims if d not in di]
+        if avail:  # change one decomposed dim; else same-decomposition case
+            do[int(rng.integers(0, m))] = int(rng.permutation(avail)[0])
+        do = tuple(do)
+        pi = tuple(rng.permutation(nd).tolist())
+        po = tuple(rng.permutation(nd).tolist())
+        dtype = [np.float64, np.float32, np.complex64][trial % 3]
+        esz = np.dtype(dtype).itemsize
+
+        topo = Topology(pdims)
+        Pi = Pencil(topo, dims, di, permute=pi)
+        Po = Pencil(topo, dims, do, permute=po)
+        g, parents = seeded_parents(dims, pdims, di, pi, (), dtype,
+                                    seed=1000 + trial)
+        from pencilarrays_amd.copyexec import apply_copy
+        for rank in range(topo.nranks):
+            plan = bp(Pi, Po, rank)
+            descs = []
+            pex = 1
+            if plan.local is not None:
+                descs.append((plan.local, parents[rank],
+                              Po.length_local(rank)))
+            for blk in plan.peers:
+                if blk.pack is not None:
+                    descs.append((blk.pack, parents[rank],
+                                  plan.send_nelem_total))
+                if blk.unpack is not None:
+                    rbuf = rng.standard_normal(
+                        plan.recv_nelem_total).astype(dtype)
+                    descs.append((blk.unpack, rbuf, Po.length_local(rank)))
+            for desc, src_np, out_n in descs:
+                if out_n == 0 or desc.nelem == 0:
+                    continue
+                exp = np.empty(out_n, dtype=dtype)
+                exp.view(np.uint8)[:] = 0xAB
+                got_t = _sentinel_like(out_n, dtype)
+                src_t = _to_gpu(np.ascontiguousarray(src_np))
+                apply_copy(desc, np.ascontiguousarray(src_np), exp)
+                dev_copy(desc, esz, src_t, got_t)
+                torch.cuda.synchronize()
+                assert np.array_equal(got_t.cpu().numpy(), exp), \
+                    f"trial {trial} rank {rank} {desc}"
+
+
 def test_rccl_single_rank_bootstrap():
     lib = native.load()
     import ctypes

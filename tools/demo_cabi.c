@@ -1,0 +1,81 @@
+/* demo_cabi.c — standalone C driver of the engine's C ABI: proves the
+ * drop-in boundary works with no Python/torch in the loop (what a Julia
+ * ccall host does).  World-1 permuted x->y transpose of a
+ * (40,30,20) Float64 grid on device memory, verified element-by-element
+ * against the definition (dest parent in Po memory order == permuted global
+ * block).
+ *
+ * Build: hipcc -x c demo_cabi.c -I../include -L../pencilarrays_amd
+ *        -lpencilhip -lamdhip64 -o demo_cabi   (run on a GPU box)
+ */
+
+#include <stdint.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+
+#include "pencilhip.h"
+
+/* minimal HIP runtime decls to avoid including hip headers from C */
+extern int hipMalloc(void **ptr, size_t size);
+extern int hipMemcpy(void *dst, const void *src, size_t size, int kind);
+extern int hipDeviceSynchronize(void);
+#define H2D 1
+#define D2H 2
+
+int main(void)
+{
+    const int64_t dims[3] = {40, 30, 20};
+    const int64_t pdims[2] = {1, 1};
+    const int32_t decomp_i[2] = {1, 2}, decomp_o[2] = {0, 2};
+    const int32_t perm_o[3] = {1, 2, 0};
+    const int64_t n = dims[0] * dims[1] * dims[2];
+
+    pa_topology *topo;
+    pa_pencil *pin, *pout;
+    pa_plan *plan;
+    if (pa_topology_create(2, pdims, &topo) ||
+        pa_pencil_create(topo, 3, dims, decomp_i, NULL, &pin) ||
+        pa_pencil_create(topo, 3, dims, decomp_o, perm_o, &pout) ||
+        pa_plan_create(pin, pout, 8, 0, NULL, 0, 0, &plan)) {
+        fprintf(stderr, "setup failed: %s\n", pa_last_error());
+        return 1;
+    }
+
+    double *h_src = malloc(n * 8), *h_dst = malloc(n * 8);
+    for (int64_t i = 0; i < n; i++) h_src[i] = (double)i; /* linear index */
+
+    void *d_src, *d_dst;
+    hipMalloc(&d_src, n * 8);
+    hipMalloc(&d_dst, n * 8);
+    hipMemcpy(d_src, h_src, n * 8, H2D);
+
+    if (pa_transpose_execute(plan, d_src, d_dst, NULL) ||
+        pa_transpose_wait(plan, NULL)) {
+        fprintf(stderr, "execute failed: %s\n", pa_last_error());
+        return 1;
+    }
+    hipMemcpy(h_dst, d_dst, n * 8, D2H);
+    hipDeviceSynchronize();
+
+    /* verify: src parent is the global array column-major (i fastest);
+     * dest parent memory order = perm (1,2,0): mem dims (30,20,40),
+     * element (j,k,i) at mem offset j + 30*k + 600*i == global (i,j,k) =
+     * i + 40*j + 1200*k. */
+    int64_t bad = 0;
+    for (int64_t i = 0; i < dims[0]; i++)
+        for (int64_t j = 0; j < dims[1]; j++)
+            for (int64_t k = 0; k < dims[2]; k++) {
+                double want = (double)(i + 40 * j + 40 * 30 * k);
+                double got = h_dst[j + 30 * k + 30 * 20 * i];
+                if (got != want) bad++;
+            }
+    printf("demo_cabi: %s (%lld mismatches of %lld)\n",
+           bad ? "FAIL" : "OK", (long long)bad, (long long)n);
+
+    pa_plan_destroy(plan);
+    pa_pencil_destroy(pin);
+    pa_pencil_destroy(pout);
+    pa_topology_destroy(topo);
+    return bad != 0;
+}

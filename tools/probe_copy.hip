@@ -168,6 +168,28 @@ __global__ __launch_bounds__(64 * NROWS) void t_nt(
                                         &dst[(j0 + j) + NJ * (i0 + i)]);
 }
 
+/* t_rect with t_j-fastest block order: consecutive blocks advance along the
+ * dst-contiguous axis, so concurrent blocks read AND write adjacent regions
+ * (DRAM page locality) on tall-skinny shapes. */
+template <int TI, int TJ, int NROWS>
+__global__ __launch_bounds__(64 * NROWS) void t_rect_jf(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t ntj)
+{
+    __shared__ uint64_t tile[TJ][TI + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_j = (int64_t)blockIdx.x % ntj;
+    const int64_t t_i = (int64_t)blockIdx.x / ntj;
+    const int64_t i0 = t_i * TI, j0 = t_j * TJ;
+    for (int j = ty; j < TJ; j += NROWS)
+        for (int i = tx; i < TI; i += 64)
+            tile[j][i] = src[(i0 + i) + NI * (j0 + j)];
+    __syncthreads();
+    for (int i = ty; i < TI; i += NROWS)
+        for (int j = tx; j < TJ; j += 64)
+            dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
+}
+
 /* vectorized 16-B loads/stores: interior tiles only (caller guarantees
  * NI,NJ multiples of TILE).  Lanes 0..31 load row 2*ty, lanes 32..63 row
  * 2*ty+1 (uint4 = 2 f64 along i).  Write phase symmetric along j. */
@@ -373,6 +395,73 @@ int main()
         TRUN2("tr rect 128x64 r16", 128, 64, 16);
         TRUN2("tr rect 256x64 r8", 256, 64, 8);
         TRUN2("tr rect 64x64 r8 (ctl)", 64, 64, 8);
+    }
+
+    /* THE REAL SHAPE: the permuted x->y unpack at 1024^3 world=1 is a
+     * (1024 x NJ) tall-skinny transpose (normalized desc dims (1024,
+     * 1048576)); probe at NJ = 262144 (2 GiB payload). */
+    {
+        const int64_t NI = 1024, NJ = 262144;
+        const double tio = 2.0 * NI * NJ * 8;
+#define TRUNS(name, kern, TI, TJ, NR, jf)                                    \
+    {                                                                        \
+        const int64_t ntX = jf ? (NJ / TJ) : (NI / TI);                      \
+        const int64_t blocks = (NI / TI) * (NJ / TJ);                        \
+        hipEvent_t a, b;                                                     \
+        CHK(hipEventCreate(&a));                                             \
+        CHK(hipEventCreate(&b));                                             \
+        hipLaunchKernelGGL((kern<TI, TJ, NR>), dim3((uint32_t)blocks),       \
+                           dim3(64, NR), 0, 0, (const uint64_t *)s,          \
+                           (uint64_t *)d, NI, NJ, ntX);                      \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < reps; r++) {                                     \
+            CHK(hipEventRecord(a));                                          \
+            hipLaunchKernelGGL((kern<TI, TJ, NR>), dim3((uint32_t)blocks),   \
+                               dim3(64, NR), 0, 0, (const uint64_t *)s,      \
+                               (uint64_t *)d, NI, NJ, ntX);                  \
+            CHK(hipEventRecord(b));                                          \
+            CHK(hipEventSynchronize(b));                                     \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a, b));                             \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("%-28s %8.1f GB/s\n", name, tio / (best * 1e-3) / 1e9);       \
+        fflush(stdout);                                                      \
+        CHK(hipEventDestroy(a));                                             \
+        CHK(hipEventDestroy(b));                                             \
+    }
+        TRUNS("RS rect 128x64 r16 (cur)", t_rect, 128, 64, 16, 0);
+        TRUNS("RS rect 64x64 r8", t_rect, 64, 64, 8, 0);
+        TRUNS("RS rect 64x128 r16", t_rect, 64, 128, 16, 0);
+        TRUNS("RS rect 32x256 r16", t_rect, 32, 256, 16, 0);
+        TRUNS("RS jf 128x64 r16", t_rect_jf, 128, 64, 16, 1);
+        TRUNS("RS jf 64x64 r8", t_rect_jf, 64, 64, 8, 1);
+        TRUNS("RS jf 64x128 r16", t_rect_jf, 64, 128, 16, 1);
+        TRUNS("RS jf 32x256 r16", t_rect_jf, 32, 256, 16, 1);
+        TRUNS("RS jf 16x512 r16", t_rect_jf, 16, 512, 16, 1);
+        /* correctness of t_rect_jf on a small pattern */
+        {
+            const int64_t ni = 256, nj = 384;
+            uint64_t *hs = (uint64_t *)malloc(ni * nj * 8);
+            uint64_t *hd = (uint64_t *)malloc(ni * nj * 8);
+            for (int64_t i = 0; i < ni * nj; i++) hs[i] = i * 2654435761ULL;
+            CHK(hipMemcpy(s, hs, ni * nj * 8, hipMemcpyHostToDevice));
+            CHK(hipMemset(d, 0xCC, ni * nj * 8));
+            hipLaunchKernelGGL((t_rect_jf<64, 64, 8>),
+                               dim3((uint32_t)((ni / 64) * (nj / 64))),
+                               dim3(64, 8), 0, 0, (const uint64_t *)s,
+                               (uint64_t *)d, ni, nj, nj / 64);
+            CHK(hipMemcpy(hd, d, ni * nj * 8, hipMemcpyDeviceToHost));
+            int64_t bad = 0;
+            for (int64_t j = 0; j < nj; j++)
+                for (int64_t i = 0; i < ni; i++)
+                    if (hd[j + nj * i] != hs[i + ni * j]) bad++;
+            printf("t_rect_jf correctness: %s (%lld bad)\n",
+                   bad ? "FAIL" : "OK", (long long)bad);
+            free(hs);
+            free(hd);
+        }
 
         /* correctness of t_vec 64x64 on a small pattern */
         {
