@@ -75,3 +75,59 @@ def test_gloo_transpose(cfg):
     port = 29511 + abs(hash(cfg)) % 2000
     mp.spawn(_worker, args=(world, pdims, dims, din, pin, dout, pout, port),
              nprocs=world, join=True)
+
+
+def _worker_inplace_chain(rank, world, port):
+    """Distributed IN-PLACE x->y->z->y->x chain over one ManyPencilArray
+    buffer per rank (the PencilFFTs in-place pattern), gloo world=4."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import sys
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        sys.path.insert(0, repo)
+        sys.path.insert(0, os.path.join(repo, "oracle"))
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        import oracle as orc
+        from pencilarrays_amd import (
+            ManyPencilArray, Pencil, Topology, Transposition,
+        )
+        from util import seeded_parents
+
+        dims, pdims = (16, 21, 41), (2, 2)
+        topo = Topology(pdims)
+        pens = (
+            Pencil(topo, dims, (1, 2)),
+            Pencil(topo, dims, (0, 2), permute=(1, 2, 0)),
+            Pencil(topo, dims, (0, 1), permute=(2, 1, 0)),
+        )
+        g, parents = seeded_parents(dims, pdims, (1, 2), (0, 1, 2), (),
+                                    np.float64)
+        m = ManyPencilArray(pens, rank)
+        n1 = pens[0].length_local(rank)
+        m.first.data[:n1] = parents[rank]
+        orig = parents[rank].copy()
+
+        t12 = Transposition(m[1], m[0])
+        assert t12.aliased
+        t12.execute()
+        t23 = Transposition(m[2], m[1])
+        t23.execute()
+        # verify z-pencil state against the oracle
+        exp = orc.transpose_oracle(parents, dims, pdims, (1, 2), (0, 1, 2),
+                                   (0, 1), (2, 1, 0), ())
+        ln = pens[2].length_local(rank)
+        assert np.array_equal(m[2].data[:ln], exp[rank]), f"rank {rank}"
+        # back down the chain, still in place
+        Transposition(m[1], m[2]).execute()
+        Transposition(m[0], m[1]).execute()
+        assert np.array_equal(m.first.data[:n1], orig)
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gloo_inplace_chain_world4():
+    mp.spawn(_worker_inplace_chain, args=(4, 29971), nprocs=4, join=True)
