@@ -440,6 +440,9 @@ int main()
         TRUNS("RS jf 64x128 r16", t_rect_jf, 64, 128, 16, 1);
         TRUNS("RS jf 32x256 r16", t_rect_jf, 32, 256, 16, 1);
         TRUNS("RS jf 16x512 r16", t_rect_jf, 16, 512, 16, 1);
+        TRUNS("RS rect 128x128 r16", t_rect, 128, 128, 16, 0);
+        TRUNS("RS rect 128x64 r8", t_rect, 128, 64, 8, 0);
+        TRUNS("RS rect 256x32 r16", t_rect, 256, 32, 16, 0);
         /* correctness of t_rect_jf on a small pattern */
         {
             const int64_t ni = 256, nj = 384;
