@@ -126,6 +126,14 @@ class Pencil:
         r = self.axes_for_rank(rank)
         return perm_apply(self.perm, r) if memory_order else r
 
+    def range_remote(self, rank_or_coords, memory_order: bool = False) -> Region:
+        """range_remote(p, coords|rank, order) (Pencils.jl:529-536)."""
+        if isinstance(rank_or_coords, tuple):
+            r = self.axes_for_coords(rank_or_coords)
+        else:
+            r = self.axes_for_rank(int(rank_or_coords))
+        return perm_apply(self.perm, r) if memory_order else r
+
     def size_local(self, rank: int, memory_order: bool = False) -> Tuple[int, ...]:
         return region_lengths(self.range_local(rank, memory_order))
 
