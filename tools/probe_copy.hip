@@ -190,6 +190,35 @@ __global__ __launch_bounds__(64 * NROWS) void t_rect_jf(
             dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
 }
 
+/* j-sweeping workgroup: each WG owns one i-block and a CONTIGUOUS span of
+ * j-tiles, so its reads advance through one contiguous src region and each
+ * of its TI output rows is written as a sequential stream (long DRAM
+ * bursts on both sides). */
+template <int TI, int TJ, int NROWS, int JCHUNK>
+__global__ __launch_bounds__(64 * NROWS) void t_sweep(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TJ][TI + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TI;
+    const int64_t jbase = chunk * (int64_t)JCHUNK * TJ;
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = jbase + (int64_t)jt * TJ;
+        if (j0 >= NJ) break;
+        for (int j = ty; j < TJ; j += NROWS)
+            for (int i = tx; i < TI; i += 64)
+                tile[j][i] = src[(i0 + i) + NI * (j0 + j)];
+        __syncthreads();
+        for (int i = ty; i < TI; i += NROWS)
+            for (int j = tx; j < TJ; j += 64)
+                dst[(j0 + j) + NJ * (i0 + i)] = tile[j][i];
+        __syncthreads();
+    }
+}
+
 /* vectorized 16-B loads/stores: interior tiles only (caller guarantees
  * NI,NJ multiples of TILE).  Lanes 0..31 load row 2*ty, lanes 32..63 row
  * 2*ty+1 (uint4 = 2 f64 along i).  Write phase symmetric along j. */
@@ -443,6 +472,67 @@ int main()
         TRUNS("RS rect 128x128 r16", t_rect, 128, 128, 16, 0);
         TRUNS("RS rect 128x64 r8", t_rect, 128, 64, 8, 0);
         TRUNS("RS rect 256x32 r16", t_rect, 256, 32, 16, 0);
+
+#define TRUNW(name, TI, TJ, NR, JC)                                          \
+    {                                                                        \
+        const int64_t nti = NI / TI;                                         \
+        const int64_t njc = (NJ / TJ + JC - 1) / JC;                         \
+        const int64_t blocks = nti * njc;                                    \
+        hipEvent_t a, b;                                                     \
+        CHK(hipEventCreate(&a));                                             \
+        CHK(hipEventCreate(&b));                                             \
+        hipLaunchKernelGGL((t_sweep<TI, TJ, NR, JC>),                        \
+                           dim3((uint32_t)blocks), dim3(64, NR), 0, 0,       \
+                           (const uint64_t *)s, (uint64_t *)d, NI, NJ, nti); \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < reps; r++) {                                     \
+            CHK(hipEventRecord(a));                                          \
+            hipLaunchKernelGGL((t_sweep<TI, TJ, NR, JC>),                    \
+                               dim3((uint32_t)blocks), dim3(64, NR), 0, 0,   \
+                               (const uint64_t *)s, (uint64_t *)d, NI, NJ,   \
+                               nti);                                         \
+            CHK(hipEventRecord(b));                                          \
+            CHK(hipEventSynchronize(b));                                     \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a, b));                             \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("%-28s %8.1f GB/s\n", name, tio / (best * 1e-3) / 1e9);       \
+        fflush(stdout);                                                      \
+        CHK(hipEventDestroy(a));                                             \
+        CHK(hipEventDestroy(b));                                             \
+    }
+        TRUNW("RS sweep 128x64 r16 c4", 128, 64, 16, 4);
+        TRUNW("RS sweep 128x64 r16 c8", 128, 64, 16, 8);
+        TRUNW("RS sweep 128x64 r16 c16", 128, 64, 16, 16);
+        TRUNW("RS sweep 128x64 r16 c32", 128, 64, 16, 32);
+        TRUNW("RS sweep 64x64 r8 c16", 64, 64, 8, 16);
+        TRUNW("RS sweep 128x128 r16 c8", 128, 128, 16, 8);
+        /* t_sweep correctness */
+        {
+            const int64_t ni = 256, nj = 448; /* non-multiple of chunk span */
+            uint64_t *hs = (uint64_t *)malloc(ni * nj * 8);
+            uint64_t *hd = (uint64_t *)malloc(ni * nj * 8);
+            for (int64_t i = 0; i < ni * nj; i++) hs[i] = i * 0x9E3779B9ULL;
+            CHK(hipMemcpy(s, hs, ni * nj * 8, hipMemcpyHostToDevice));
+            CHK(hipMemset(d, 0xCC, ni * nj * 8));
+            const int64_t nti_ = ni / 64;
+            const int64_t njc_ = (nj / 64 + 3) / 4;
+            hipLaunchKernelGGL((t_sweep<64, 64, 8, 4>),
+                               dim3((uint32_t)(nti_ * njc_)), dim3(64, 8), 0,
+                               0, (const uint64_t *)s, (uint64_t *)d, ni, nj,
+                               nti_);
+            CHK(hipMemcpy(hd, d, ni * nj * 8, hipMemcpyDeviceToHost));
+            int64_t bad = 0;
+            for (int64_t j = 0; j < nj; j++)
+                for (int64_t i = 0; i < ni; i++)
+                    if (hd[j + nj * i] != hs[i + ni * j]) bad++;
+            printf("t_sweep correctness: %s (%lld bad)\n", bad ? "FAIL" : "OK",
+                   (long long)bad);
+            free(hs);
+            free(hd);
+        }
         /* correctness of t_rect_jf on a small pattern */
         {
             const int64_t ni = 256, nj = 384;
