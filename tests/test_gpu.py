@@ -321,6 +321,14 @@ def test_rccl_single_rank_bootstrap():
     assert lib.pa_get_unique_id(buf) == 0, lib.pa_last_error().decode()
     comm = native.NativeComm.create(bytes(buf.raw), 1, 0)
     assert comm.handle
+    # pa_allreduce through RCCL (world=1: identity), f64 sum in place
+    t = torch.arange(1024, dtype=torch.float64, device="cuda:0")
+    st = lib.pa_allreduce(comm.handle, ctypes.c_void_p(t.data_ptr()),
+                          ctypes.c_void_p(t.data_ptr()),
+                          ctypes.c_int64(1024), 0, 0, None)
+    assert st == 0, lib.pa_last_error().decode()
+    torch.cuda.synchronize()
+    assert torch.equal(t.cpu(), torch.arange(1024, dtype=torch.float64))
 
 
 def test_roundtrip_on_gpu():
