@@ -168,35 +168,48 @@ class NativeComm:
 _COMM_CACHE: Dict[tuple, NativeComm] = {}
 
 
+def exchange_uid(topology, r_dim: int, rank: int, uid_fn) -> Tuple[bytes, int, int]:
+    """Exchange the subgroup's RCCL unique id through the torch.distributed
+    store: the subgroup leader (coordinate 0 along r_dim) generates it with
+    ``uid_fn()`` and publishes it under a key unique to the subgroup; the
+    others fetch it.  Returns (uid, subgroup_size, subgroup_rank).
+    Assumes dist rank == topology rank (one process per GPU)."""
+    ranks = topology.subgroup_ranks(rank, r_dim)
+    import torch.distributed as dist
+    if not (dist.is_available() and dist.is_initialized()):
+        raise RuntimeError(
+            "multi-GPU transpose needs torch.distributed initialised for the "
+            "RCCL unique-id exchange")
+    store = dist.distributed_c10d._get_default_store()
+    sub_rank = ranks.index(rank)
+    store_key = f"pencilhip_uid_{topology.dims}_{r_dim}_{min(ranks)}"
+    if sub_rank == 0:
+        uid = uid_fn()
+        store.set(store_key, uid)
+    else:
+        uid = bytes(store.get(store_key))
+    return uid, len(ranks), sub_rank
+
+
+def _nccl_uid() -> bytes:
+    lib = load()
+    n = lib.pa_unique_id_size()
+    buf = ctypes.create_string_buffer(n)
+    _check(lib, lib.pa_get_unique_id(buf), "pa_get_unique_id")
+    return bytes(buf.raw)
+
+
 def subgroup_comm(topology, r_dim: int, rank: int) -> Optional[NativeComm]:
     """RCCL communicator for the 1-D subgroup through ``rank`` along
-    ``r_dim``; unique id exchanged via the torch.distributed store.
-    Assumes dist rank == topology rank (one process per GPU)."""
+    ``r_dim`` (replacing topology.subcomms[R], MPITopologies.jl:244-251)."""
     ranks = topology.subgroup_ranks(rank, r_dim)
     if len(ranks) == 1:
         return None
     key = (tuple(topology.dims), r_dim, tuple(ranks))
     if key in _COMM_CACHE:
         return _COMM_CACHE[key]
-
-    import torch.distributed as dist
-    if not (dist.is_available() and dist.is_initialized()):
-        raise RuntimeError(
-            "multi-GPU transpose needs torch.distributed initialised for the "
-            "RCCL unique-id exchange")
-    lib = load()
-    store = dist.distributed_c10d._get_default_store()
-    sub_rank = ranks.index(rank)
-    store_key = f"pencilhip_uid_{topology.dims}_{r_dim}_{min(ranks)}"
-    if sub_rank == 0:
-        n = lib.pa_unique_id_size()
-        buf = ctypes.create_string_buffer(n)
-        _check(lib, lib.pa_get_unique_id(buf), "pa_get_unique_id")
-        store.set(store_key, bytes(buf.raw))
-        uid = bytes(buf.raw)
-    else:
-        uid = bytes(store.get(store_key))
-    comm = NativeComm.create(uid, len(ranks), sub_rank)
+    uid, nranks, sub_rank = exchange_uid(topology, r_dim, rank, _nccl_uid)
+    comm = NativeComm.create(uid, nranks, sub_rank)
     _COMM_CACHE[key] = comm
     return comm
 

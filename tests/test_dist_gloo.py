@@ -131,3 +131,41 @@ def _worker_inplace_chain(rank, world, port):
 
 def test_gloo_inplace_chain_world4():
     mp.spawn(_worker_inplace_chain, args=(4, 29971), nprocs=4, join=True)
+
+
+def _worker_uid_exchange(rank, world, port):
+    """The RCCL unique-id bootstrap (native.exchange_uid) with an injected
+    uid generator: every subgroup member must receive its leader's 128-byte
+    id, with per-subgroup keys and coordinate-ordered subgroup ranks —
+    exercised cross-process on gloo (the NCCL call itself needs a GPU)."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import sys
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        sys.path.insert(0, repo)
+        from pencilarrays_amd import Topology
+        from pencilarrays_amd.native import exchange_uid
+
+        topo = Topology((2, 2))
+        for r_dim in (0, 1):
+            ranks = topo.subgroup_ranks(rank, r_dim)
+            leader = ranks[0]
+            fake = bytes([leader]) * 128  # distinct per subgroup leader
+
+            uid, nranks, sub_rank = exchange_uid(
+                topo, r_dim, rank, uid_fn=lambda: fake)
+            assert nranks == 2
+            assert sub_rank == ranks.index(rank)
+            assert sub_rank == topo.cart_coords(rank)[r_dim]
+            assert uid == bytes([leader]) * 128, \
+                f"rank {rank} dim {r_dim}: wrong uid"
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gloo_uid_exchange_world4():
+    mp.spawn(_worker_uid_exchange, args=(4, 29181), nprocs=4, join=True)
