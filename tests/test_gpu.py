@@ -331,6 +331,48 @@ def test_rccl_single_rank_bootstrap():
     assert torch.equal(t.cpu(), torch.arange(1024, dtype=torch.float64))
 
 
+def test_fft_pattern_on_gpu():
+    """PencilFFTs consumer pattern on device: per-axis torch.fft on the
+    contiguous memory axis of each pencil + native transposes, vs
+    torch.fft.fftn of the whole array (world=1, permuted pencils)."""
+    from pencilarrays_amd.permutations import perm_inv
+    dims = (64, 48, 40)
+    topo = Topology((1, 1))
+    p1 = Pencil(topo, dims, (1, 2))
+    p2 = Pencil(topo, dims, (0, 2), permute=(1, 2, 0))
+    p3 = Pencil(topo, dims, (0, 1), permute=(2, 1, 0))
+
+    rng = np.random.default_rng(5)
+    g = (rng.standard_normal(dims) + 1j * rng.standard_normal(dims)
+         ).astype(np.complex128)
+    parent = np.asfortranarray(g).ravel(order="F")
+
+    u1 = PencilArray(p1, 0, _to_gpu(parent))
+    u2 = PencilArray(p2, 0, _sentinel_like(p2.length_local(0), np.complex128))
+    u3 = PencilArray(p3, 0, _sentinel_like(p3.length_local(0), np.complex128))
+
+    def fft_axis0(x):
+        mv = x.parent_memview()  # torch view, axis 0 fastest
+        assert perm_inv(x.pencil.perm)[fft_axis0.dim] == 0
+        out = torch.fft.fft(mv, dim=0)
+        mv.copy_(out)
+
+    fft_axis0.dim = 0
+    fft_axis0(u1)
+    Transposition(u2, u1).execute()
+    fft_axis0.dim = 1
+    fft_axis0(u2)
+    Transposition(u3, u2).execute()
+    fft_axis0.dim = 2
+    fft_axis0(u3)
+    torch.cuda.synchronize()
+
+    # u3 parent (mem order (2,1,0)) -> logical order
+    got = u3.logical_view().cpu().numpy()
+    want = np.fft.fftn(g)
+    assert np.allclose(got, want, rtol=1e-10, atol=1e-7)
+
+
 def test_roundtrip_on_gpu():
     """u1 -> u2 -> u1 world=1 with permuted pencils restores u1 bit-exactly
     (test/transpose.jl:48-60 recipe)."""
