@@ -52,6 +52,10 @@ def parse_args():
                          "transpose (PencilFFTs pattern), deferred waits")
     ap.add_argument("--dtype", default="float64",
                     choices=["float64", "complex64"])
+    ap.add_argument("--chunks", type=int, default=1,
+                    help="split the RCCL exchange into N chunks and unpack "
+                         "each as it lands (the reference's Waitany overlap; "
+                         "1 = single grouped exchange)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     return ap.parse_args()
 
@@ -87,6 +91,8 @@ def cpu_baseline(sample=(512, 512, 512), reps=3):
 
 def main():
     args = parse_args()
+    if args.chunks > 1:
+        os.environ["PENCILHIP_EXCHANGE_CHUNKS"] = str(args.chunks)
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))

@@ -522,6 +522,13 @@ struct PeerBlockC {
     int64_t send_off, recv_off, send_n, recv_n;
     bool has_pack = false, has_unpack = false;
     CopyDescH pack, unpack;
+    /* chunked-exchange support: the send/recv blocks viewed as
+     * (rowelems x outer) with outer = the last raw (Pi-mem-order + extras)
+     * axis — identical split on sender and receiver by construction. */
+    int64_t send_outer = 1, send_rowelems = 0;
+    int64_t recv_outer = 1, recv_rowelems = 0;
+    CopyDescH unpack_raw; /* un-normalized: nd = n+E, axis nd-1 = outer */
+    bool has_unpack_raw = false;
 };
 
 struct pa_plan {
@@ -551,6 +558,12 @@ struct pa_plan {
      * unpacks it while recvs are in flight, Transpositions.jl:510-517). */
     hipStream_t comm_stream = nullptr;
     hipEvent_t ev_pack = nullptr, ev_comm = nullptr;
+    /* >1: split the exchange into this many chunks and unpack each chunk
+     * while later chunks are still in flight (the reference's Waitany
+     * overlap, Transpositions.jl:510-517).  From PENCILHIP_EXCHANGE_CHUNKS;
+     * default 1 (single grouped exchange). */
+    int chunks = 1;
+    std::vector<hipEvent_t> ev_chunks;
 };
 
 static int64_t prod_extra(const pa_plan &pl)
@@ -607,9 +620,22 @@ static void window_desc(const pa_pencil &p, int rank,
  * column-major with given strides; dest axis i' reads buffer axis
  * inv(perm_i)[perm_o[i']] (the relative permutation, Transpositions.jl:506).
  */
+static CopyDescH unpack_desc_raw_out(const pa_plan &pl, const Range *grange,
+                                     const int64_t *bufdims,
+                                     const int64_t *bufstr, int64_t bufoff,
+                                     CopyDescH *raw_out);
+
 static CopyDescH unpack_desc(const pa_plan &pl, const Range *grange,
                              const int64_t *bufdims, const int64_t *bufstr,
                              int64_t bufoff)
+{
+    return unpack_desc_raw_out(pl, grange, bufdims, bufstr, bufoff, nullptr);
+}
+
+static CopyDescH unpack_desc_raw_out(const pa_plan &pl, const Range *grange,
+                                     const int64_t *bufdims,
+                                     const int64_t *bufstr, int64_t bufoff,
+                                     CopyDescH *raw_out)
 {
     const pa_pencil &Po = pl.Po;
     const pa_pencil &Pi = pl.Pi;
@@ -633,6 +659,18 @@ static CopyDescH unpack_desc(const pa_plan &pl, const Range *grange,
     }
     for (int e = 0; e < E; e++) dstr[n + e] = pst_o[n + e];
 
+    if (raw_out) {
+        raw_out->nd = n + E;
+        for (int i = 0; i < n + E; i++) {
+            raw_out->dims[i] = bufdims[i];
+            raw_out->sstr[i] = bufstr[i];
+            raw_out->dstr[i] = dstr[i];
+        }
+        raw_out->soff = bufoff;
+        raw_out->doff = doff;
+        raw_out->total = 1;
+        for (int i = 0; i < n + E; i++) raw_out->total *= bufdims[i];
+    }
     return normalize_desc(n + E, bufdims, bufstr, bufoff, dstr, doff);
 }
 
@@ -665,6 +703,28 @@ static void staged_self_descs(pa_plan *pl, const Range *sr, const Range *rr,
     }
     pl->self_unpack = unpack_desc(*pl, rr, bdims, bstr, recv_off);
     pl->has_self = true;
+}
+
+
+/* chunk of a raw (un-normalized) unpack descriptor: outer rows [lo, hi) of
+ * its last axis */
+static CopyDescH chunk_of_raw(const CopyDescH &raw, int64_t lo, int64_t hi)
+{
+    const int last = raw.nd - 1;
+    int64_t dims[2 * MAXND], sstr[2 * MAXND], dstr[2 * MAXND];
+    for (int i = 0; i < raw.nd; i++) {
+        dims[i] = raw.dims[i];
+        sstr[i] = raw.sstr[i];
+        dstr[i] = raw.dstr[i];
+    }
+    dims[last] = hi - lo;
+    return normalize_desc(raw.nd, dims, sstr, raw.soff + lo * sstr[last],
+                          dstr, raw.doff + lo * dstr[last]);
+}
+
+static inline int64_t chunk_lo(int64_t outer, int c, int C)
+{
+    return outer * c / C;
 }
 
 extern "C" {
@@ -826,7 +886,8 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                     "dimension.");
     if (rank < 0 || rank >= pin->topo.nranks) return fail("rank out of range");
     if (elem_size <= 0) return fail("invalid elem_size");
-    if (pin->n + e + 1 > 2 * MAXND) return fail("too many dims");
+    if (pin->n + e > MAXND)
+        return fail("too many dims: N+E must be <= %d", MAXND);
 
     auto *pl = new pa_plan;
     pl->Pi = *pin;
@@ -837,6 +898,10 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
     pl->esz = elem_size;
     pl->R = R;
     pl->aliased = (flags & 1) != 0;
+    if (const char *ce = getenv("PENCILHIP_EXCHANGE_CHUNKS")) {
+        int c = atoi(ce);
+        if (c >= 1 && c <= 64) pl->chunks = c;
+    }
 
     const int n = pin->n;
     const int64_t pex = prod_extra(*pl);
@@ -929,6 +994,8 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                 }
                 blk.pack = normalize_desc(kk, dims, str, off, cstr, isend);
                 blk.has_pack = true;
+                blk.send_outer = dims[kk - 1];
+                blk.send_rowelems = blk.send_n / (dims[kk - 1] ? dims[kk - 1] : 1);
             }
             if (blk.recv_n > 0) {
                 /* buffer dims: recv block extents gathered by Pi's
@@ -942,8 +1009,13 @@ pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                     bstr[i] = acc;
                     acc *= bdims[i];
                 }
-                blk.unpack = unpack_desc(*pl, rr, bdims, bstr, irecv);
+                blk.unpack = unpack_desc_raw_out(*pl, rr, bdims, bstr,
+                                                 irecv, &blk.unpack_raw);
                 blk.has_unpack = true;
+                blk.has_unpack_raw = true;
+                blk.recv_outer = bdims[(n + e) - 1];
+                blk.recv_rowelems =
+                    blk.recv_n / (blk.recv_outer ? blk.recv_outer : 1);
             }
             isend += blk.send_n;
             irecv += blk.recv_n;
@@ -982,6 +1054,7 @@ void pa_plan_destroy(pa_plan *p)
     }
     if (p->ev_pack) (void)hipEventDestroy(p->ev_pack);
     if (p->ev_comm) (void)hipEventDestroy(p->ev_comm);
+    for (auto e : p->ev_chunks) (void)hipEventDestroy(e);
     if (p->comm_stream) (void)hipStreamDestroy(p->comm_stream);
     delete p;
 }
@@ -1061,6 +1134,7 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
             if (blk.k != p->myk && (blk.send_n || blk.recv_n))
                 exchanging = true;
     }
+    const int C = (exchanging && p->chunks > 1) ? p->chunks : 1;
     if (exchanging) {
         if (!p->comm)
             return fail("subgroup exchange requires pa_plan_set_comm");
@@ -1072,22 +1146,48 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
             HIP_CHECK(hipEventCreateWithFlags(&p->ev_comm,
                                               hipEventDisableTiming));
         }
+        while ((int)p->ev_chunks.size() < C) {
+            hipEvent_t e;
+            HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+            p->ev_chunks.push_back(e);
+        }
         HIP_CHECK(hipEventRecord(p->ev_pack, stream));
         HIP_CHECK(hipStreamWaitEvent(p->comm_stream, p->ev_pack, 0));
-        NCCL_CHECK(ncclGroupStart());
-        for (auto &blk : p->peers) {
-            if (blk.k == p->myk) continue;
-            if (blk.recv_n)
-                NCCL_CHECK(ncclRecv((char *)p->recv_buf + blk.recv_off * p->esz,
-                                    (size_t)(blk.recv_n * p->esz), ncclUint8,
-                                    blk.k, p->comm->comm, p->comm_stream));
-            if (blk.send_n)
-                NCCL_CHECK(ncclSend((const char *)p->send_buf +
-                                        blk.send_off * p->esz,
-                                    (size_t)(blk.send_n * p->esz), ncclUint8,
-                                    blk.k, p->comm->comm, p->comm_stream));
+        /* C == 1: one grouped exchange.  C > 1: the exchange is split into C
+         * groups of matching per-peer sub-blocks (outer rows of the block,
+         * identical split on sender and receiver), and each chunk's unpack
+         * overlaps the later chunks' transfers — the reference's Waitany
+         * overlap (Transpositions.jl:510-517) in stream/event form. */
+        for (int c = 0; c < C; c++) {
+            NCCL_CHECK(ncclGroupStart());
+            for (auto &blk : p->peers) {
+                if (blk.k == p->myk) continue;
+                if (blk.recv_n) {
+                    const int64_t lo = chunk_lo(blk.recv_outer, c, C);
+                    const int64_t hi = chunk_lo(blk.recv_outer, c + 1, C);
+                    if (hi > lo)
+                        NCCL_CHECK(ncclRecv(
+                            (char *)p->recv_buf +
+                                (blk.recv_off + lo * blk.recv_rowelems) *
+                                    p->esz,
+                            (size_t)((hi - lo) * blk.recv_rowelems * p->esz),
+                            ncclUint8, blk.k, p->comm->comm, p->comm_stream));
+                }
+                if (blk.send_n) {
+                    const int64_t lo = chunk_lo(blk.send_outer, c, C);
+                    const int64_t hi = chunk_lo(blk.send_outer, c + 1, C);
+                    if (hi > lo)
+                        NCCL_CHECK(ncclSend(
+                            (const char *)p->send_buf +
+                                (blk.send_off + lo * blk.send_rowelems) *
+                                    p->esz,
+                            (size_t)((hi - lo) * blk.send_rowelems * p->esz),
+                            ncclUint8, blk.k, p->comm->comm, p->comm_stream));
+                }
+            }
+            NCCL_CHECK(ncclGroupEnd());
+            HIP_CHECK(hipEventRecord(p->ev_chunks[c], p->comm_stream));
         }
-        NCCL_CHECK(ncclGroupEnd());
         HIP_CHECK(hipEventRecord(p->ev_comm, p->comm_stream));
     }
 
@@ -1104,15 +1204,33 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
         if (st) return st;
     }
 
-    /* 4. unpack every received block (:489-536) after the exchange lands */
-    if (exchanging)
-        HIP_CHECK(hipStreamWaitEvent(stream, p->ev_comm, 0));
-    for (auto &blk : p->peers)
-        if (blk.has_unpack) {
-            pa_status st = launch_desc(blk.unpack, p->esz, p->recv_buf,
-                                       dst_parent, stream);
-            if (st) return st;
+    /* 4. unpack received blocks (:489-536).  C == 1: all after the single
+     * exchange completes.  C > 1: chunk c unpacks as soon as its group
+     * lands, overlapping chunks c+1.. in flight. */
+    if (C <= 1) {
+        if (exchanging)
+            HIP_CHECK(hipStreamWaitEvent(stream, p->ev_comm, 0));
+        for (auto &blk : p->peers)
+            if (blk.has_unpack) {
+                pa_status st = launch_desc(blk.unpack, p->esz, p->recv_buf,
+                                           dst_parent, stream);
+                if (st) return st;
+            }
+    } else {
+        for (int c = 0; c < C; c++) {
+            HIP_CHECK(hipStreamWaitEvent(stream, p->ev_chunks[c], 0));
+            for (auto &blk : p->peers) {
+                if (!blk.has_unpack_raw) continue;
+                const int64_t lo = chunk_lo(blk.recv_outer, c, C);
+                const int64_t hi = chunk_lo(blk.recv_outer, c + 1, C);
+                if (hi <= lo) continue;
+                CopyDescH d = chunk_of_raw(blk.unpack_raw, lo, hi);
+                pa_status st =
+                    launch_desc(d, p->esz, p->recv_buf, dst_parent, stream);
+                if (st) return st;
+            }
         }
+    }
 
     return 0;
 }
@@ -1160,6 +1278,13 @@ pa_status pa_plan_copydesc(const pa_plan *p, int which, int k, int64_t *nd,
     } else if (which == 4) {
         if (!p->has_self) return fail("no staged self unpack in plan");
         d = &p->self_unpack;
+    } else if (which == 5) {
+        if (p->R < 0 || k < 0 || k >= (int)p->peers.size())
+            return fail("no peer block %d", k);
+        if (!p->peers[k].has_unpack_raw) return fail("no raw unpack");
+        d = &p->peers[k].unpack_raw;
+        /* raw: may exceed normalized nd cap assumptions of callers; nd <= 2*MAXND
+         * but CopyDescH holds MAXND — enforced at plan build (n+E <= MAXND). */
     } else {
         if (p->R < 0 || k < 0 || k >= (int)p->peers.size())
             return fail("no peer block %d", k);
