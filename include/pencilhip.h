@@ -84,7 +84,11 @@ void pa_comm_destroy(pa_comm *c);
  * flags bit 0 (PA_PLAN_ALIASED): src and dst parents may alias (the
  * reference's in-place / ManyPencilArray transposes, multiarrays.jl:106-143,
  * Transpositions.jl:250-264): the self block then stages through the
- * recv-buffer tail so every src read completes before any dst write. */
+ * recv-buffer tail so every src read completes before any dst write.
+ * Env PENCILHIP_EXCHANGE_CHUNKS=N (read at plan creation, default 1):
+ * split the exchange into N chunk groups and unpack each chunk while later
+ * chunks are in flight (the reference's Waitany overlap,
+ * Transpositions.jl:510-517). */
 #define PA_PLAN_ALIASED 1
 pa_status pa_plan_create(const pa_pencil *pin, const pa_pencil *pout,
                          int64_t elem_size, int e, const int64_t *extra_dims,
