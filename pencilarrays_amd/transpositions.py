@@ -155,7 +155,12 @@ class Transposition:
         the call returns with work enqueued on the stream; call :meth:`wait`
         (== ``MPI.Waitall(t)``, :128-131) before reusing the source or
         relying on the destination outside the stream."""
-        if self.src.is_torch and self.src.data.is_cuda:
+        if self.src.is_torch:
+            if not self.src.data.is_cuda:
+                raise RuntimeError(
+                    "torch CPU tensors are not a supported backend: use "
+                    "numpy arrays for the CPU host mirror, or cuda tensors "
+                    "for the native engine")
             self._execute_torch_cuda(sync)
         else:
             import torch.distributed as dist
