@@ -122,12 +122,9 @@ def main():
     Po = Pencil(topo, dims, (0, 2), permute=po_perm)
     Pz = Pencil(topo, dims, (0, 1), permute=(2, 1, 0)) if args.double else None
 
-    gen = torch.Generator(device="cpu").manual_seed(0xC0FFEE + rank)
+    gen = torch.Generator(device=device).manual_seed(0xC0FFEE + rank)
     n_in = Pi.length_local(rank)
-    if tdt.is_complex:
-        src_t = torch.randn(n_in, generator=gen, dtype=tdt).to(device)
-    else:
-        src_t = torch.randn(n_in, generator=gen, dtype=tdt).to(device)
+    src_t = torch.randn(n_in, generator=gen, dtype=tdt, device=device)
     src = PencilArray(Pi, rank, src_t)
     dst = PencilArray(
         Po, rank, torch.empty(Po.length_local(rank), dtype=tdt, device=device))
