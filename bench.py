@@ -124,7 +124,14 @@ def main():
 
     gen = torch.Generator(device=device).manual_seed(0xC0FFEE + rank)
     n_in = Pi.length_local(rank)
-    src_t = torch.randn(n_in, generator=gen, dtype=tdt, device=device)
+    src_t = torch.empty(n_in, dtype=tdt, device=device)
+    # fill in <=2^30-element chunks (some torch fill kernels reject >2^31
+    # element launches; our own kernels are 64-bit-indexed throughout)
+    CH = 1 << 30
+    for off in range(0, n_in, CH):
+        sl = src_t[off:off + CH]
+        sl.copy_(torch.randn(sl.numel(), generator=gen, dtype=tdt,
+                             device=device))
     src = PencilArray(Pi, rank, src_t)
     dst = PencilArray(
         Po, rank, torch.empty(Po.length_local(rank), dtype=tdt, device=device))
