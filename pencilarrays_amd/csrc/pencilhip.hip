@@ -95,7 +95,11 @@ __global__ __launch_bounds__(256) void k_copy_1d(const T *__restrict__ src,
                                                  T *__restrict__ dst,
                                                  int64_t n)
 {
-    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    /* 2-D grid: the HSA dispatch packet's per-dimension work-item count is
+     * 32-bit, so >2^32 threads (e.g. 2048^3 f64 = exactly 2^32 uint4) must
+     * split across gridDim.y. */
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
     if (i < n) dst[i] = src[i];
 }
 
@@ -108,7 +112,8 @@ __global__ __launch_bounds__(256) void k_copy_linear(const T *__restrict__ src,
                                                      DescDev d)
 {
     const int64_t run = d.dims[0];
-    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t idx = b * blockDim.x + threadIdx.x;
     if (idx >= d.total) return;
     int64_t o = idx / run;
     const int64_t i = idx - o * run;
@@ -129,8 +134,9 @@ __global__ __launch_bounds__(256) void k_copy_generic(const T *__restrict__ src,
                                                       T *__restrict__ dst,
                                                       DescDev d)
 {
-    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t idx = ((int64_t)blockIdx.y * gridDim.x + blockIdx.x) *
+                      blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * gridDim.y * blockDim.x;
     for (; idx < d.total; idx += stride) {
         int64_t rem = idx, so = d.soff, doo = d.doff;
         for (int a = 0; a < d.nd; a++) {
@@ -155,7 +161,7 @@ __global__ __launch_bounds__(256) void k_copy_generic(const T *__restrict__ src,
 template <typename T, int TILE_I, int TILE_J, int NROWS>
 __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
     const T *__restrict__ src, T *__restrict__ dst, DescDev d, int ta,
-    int64_t ntile_i, int64_t ntile_j)
+    int64_t ntile_i, int64_t ntile_j, int64_t nblocks)
 {
     __shared__ T tile[TILE_J][TILE_I + 1];
 
@@ -164,7 +170,8 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
 
     const int64_t tiles_per_batch = ntile_i * ntile_j;
 
-    int64_t bid = blockIdx.x;
+    int64_t bid = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    if (bid >= nblocks) return;
     const int64_t tij = bid % tiles_per_batch;
     int64_t batch = bid / tiles_per_batch;
     const int64_t t_i = tij % ntile_i;
@@ -293,6 +300,19 @@ static int64_t grid_exact(int64_t work_items, int per_block)
     return blocks;
 }
 
+/* Split a block count into a 2-D grid: the HSA dispatch limit is 2^32-1
+ * WORK-ITEMS per grid dimension, so gridDim.x is capped at that / threads
+ * and the remainder goes to gridDim.y. */
+static pa_status grid2d(int64_t blocks, int threads_per_block, dim3 *out)
+{
+    const int64_t max_x = 0xFFFFFFFFll / threads_per_block;
+    int64_t gx = blocks < max_x ? blocks : max_x;
+    int64_t gy = (blocks + gx - 1) / gx;
+    if (gy > 0xFFFFFFFFll) return fail("grid too large");
+    *out = dim3((uint32_t)gx, (uint32_t)gy);
+    return 0;
+}
+
 static int grid_for(int64_t work_items, int per_block)
 {
     int64_t blocks = grid_exact(work_items, per_block);
@@ -320,46 +340,46 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
             W = 1;
         }
         if (w.nd == 1) {
-            const int64_t blocks64 = grid_exact(w.total, 256);
-            if (blocks64 > 0x7FFFFFFF) return fail("copy grid too large");
-            const uint32_t blocks = (uint32_t)blocks64;
+            dim3 blocks;
+            pa_status gst = grid2d(grid_exact(w.total, 256), 256, &blocks);
+            if (gst) return gst;
             if (W == 16)
-                hipLaunchKernelGGL(k_copy_1d<uint4>, dim3(blocks), dim3(256),
+                hipLaunchKernelGGL(k_copy_1d<uint4>, blocks, dim3(256),
                                    0, stream, (const uint4 *)s + w.soff,
                                    (uint4 *)d + w.doff, w.total);
             else if (W == 8)
-                hipLaunchKernelGGL(k_copy_1d<uint64_t>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_1d<uint64_t>, blocks,
                                    dim3(256), 0, stream,
                                    (const uint64_t *)s + w.soff,
                                    (uint64_t *)d + w.doff, w.total);
             else if (W == 4)
-                hipLaunchKernelGGL(k_copy_1d<uint32_t>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_1d<uint32_t>, blocks,
                                    dim3(256), 0, stream,
                                    (const uint32_t *)s + w.soff,
                                    (uint32_t *)d + w.doff, w.total);
             else
-                hipLaunchKernelGGL(k_copy_1d<uint8_t>, dim3(blocks), dim3(256),
+                hipLaunchKernelGGL(k_copy_1d<uint8_t>, blocks, dim3(256),
                                    0, stream, (const uint8_t *)s + w.soff,
                                    (uint8_t *)d + w.doff, w.total);
         } else {
             DescDev dd = to_dev(w);
-            const int64_t blocks64 = grid_exact(w.total, 256);
-            if (blocks64 > 0x7FFFFFFF) return fail("copy grid too large");
-            const uint32_t blocks = (uint32_t)blocks64;
+            dim3 blocks;
+            pa_status gst = grid2d(grid_exact(w.total, 256), 256, &blocks);
+            if (gst) return gst;
             if (W == 16)
-                hipLaunchKernelGGL(k_copy_linear<uint4>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_linear<uint4>, blocks,
                                    dim3(256), 0, stream, (const uint4 *)s,
                                    (uint4 *)d, dd);
             else if (W == 8)
-                hipLaunchKernelGGL(k_copy_linear<uint64_t>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_linear<uint64_t>, blocks,
                                    dim3(256), 0, stream, (const uint64_t *)s,
                                    (uint64_t *)d, dd);
             else if (W == 4)
-                hipLaunchKernelGGL(k_copy_linear<uint32_t>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_linear<uint32_t>, blocks,
                                    dim3(256), 0, stream, (const uint32_t *)s,
                                    (uint32_t *)d, dd);
             else
-                hipLaunchKernelGGL(k_copy_linear<uint8_t>, dim3(blocks),
+                hipLaunchKernelGGL(k_copy_linear<uint8_t>, blocks,
                                    dim3(256), 0, stream, (const uint8_t *)s,
                                    (uint8_t *)d, dd);
         }
@@ -385,28 +405,32 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
             constexpr int TI = 128, TJ = 64, NR = 16;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
-            const int64_t blocks = nti * ntj * nbatch;
-            if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
+            const int64_t nblocks = nti * ntj * nbatch;
+            dim3 grid;
+            pa_status gst = grid2d(nblocks, 64 * NR, &grid);
+            if (gst) return gst;
             if (esz == 8)
                 hipLaunchKernelGGL((k_transpose_tile<uint64_t, TI, TJ, NR>),
-                                   dim3((uint32_t)blocks), dim3(64, NR), 0,
+                                   grid, dim3(64, NR), 0,
                                    stream, (const uint64_t *)s, (uint64_t *)d,
-                                   dd, ta, nti, ntj);
+                                   dd, ta, nti, ntj, nblocks);
             else
                 hipLaunchKernelGGL((k_transpose_tile<uint32_t, TI, TJ, NR>),
-                                   dim3((uint32_t)blocks), dim3(64, NR), 0,
+                                   grid, dim3(64, NR), 0,
                                    stream, (const uint32_t *)s, (uint32_t *)d,
-                                   dd, ta, nti, ntj);
+                                   dd, ta, nti, ntj, nblocks);
         } else {
             constexpr int TI = 32, TJ = 32, NR = 8;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
-            const int64_t blocks = nti * ntj * nbatch;
-            if (blocks > 0x7FFFFFFF) return fail("transpose grid too large");
+            const int64_t nblocks = nti * ntj * nbatch;
+            dim3 grid;
+            pa_status gst = grid2d(nblocks, 64 * NR, &grid);
+            if (gst) return gst;
             hipLaunchKernelGGL((k_transpose_tile<uint4, TI, TJ, NR>),
-                               dim3((uint32_t)blocks), dim3(64, NR), 0,
+                               grid, dim3(64, NR), 0,
                                stream, (const uint4 *)s, (uint4 *)d, dd, ta,
-                               nti, ntj);
+                               nti, ntj, nblocks);
         }
         HIP_CHECK(hipGetLastError());
         return 0;
@@ -415,16 +439,16 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
     /* generic fallback */
     {
         DescDev dd = to_dev(dn);
-        const int blocks = grid_for(dn.total, 256);
+        const dim3 blocks(grid_for(dn.total, 256)); /* grid-stride kernel */
         if (esz == 16)
-            hipLaunchKernelGGL(k_copy_generic<uint4>, dim3(blocks), dim3(256),
+            hipLaunchKernelGGL(k_copy_generic<uint4>, blocks, dim3(256),
                                0, stream, (const uint4 *)s, (uint4 *)d, dd);
         else if (esz == 8)
-            hipLaunchKernelGGL(k_copy_generic<uint64_t>, dim3(blocks),
+            hipLaunchKernelGGL(k_copy_generic<uint64_t>, blocks,
                                dim3(256), 0, stream, (const uint64_t *)s,
                                (uint64_t *)d, dd);
         else if (esz == 4)
-            hipLaunchKernelGGL(k_copy_generic<uint32_t>, dim3(blocks),
+            hipLaunchKernelGGL(k_copy_generic<uint32_t>, blocks,
                                dim3(256), 0, stream, (const uint32_t *)s,
                                (uint32_t *)d, dd);
         else

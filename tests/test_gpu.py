@@ -381,6 +381,33 @@ def test_randomized_descriptor_sweep():
                     f"trial {trial} rank {rank} {desc}"
 
 
+def test_launch_beyond_2e32_workitems():
+    """Regression: the HSA dispatch limit is 2^32-1 work-items per grid
+    dimension; launches above it must split across gridDim.y (a 2048^3 f64
+    copy is exactly 2^32 16-B words and used to fail).  Cheap variant: a
+    byte-element copy of 2^32+13 bytes (odd, so no word-scaling) needs
+    ~4.3e9 threads."""
+    import ctypes
+    lib = native.load()
+    I64 = ctypes.c_int64
+    n = (1 << 32) + 13
+    src = torch.empty(n, dtype=torch.uint8, device="cuda:0")
+    src[:256].fill_(7)
+    src[-256:].fill_(9)
+    dst = torch.zeros(n, dtype=torch.uint8, device="cuda:0")
+    st = lib.pa_device_copy(
+        1, (I64 * 1)(n), (I64 * 1)(1), I64(0), (I64 * 1)(1), I64(0),
+        I64(1), ctypes.c_void_p(src.data_ptr()),
+        ctypes.c_void_p(dst.data_ptr()), None)
+    assert st == 0, lib.pa_last_error().decode()
+    torch.cuda.synchronize()
+    assert torch.equal(dst[:256].cpu(), src[:256].cpu())
+    assert torch.equal(dst[-256:].cpu(), src[-256:].cpu())
+    assert int(dst[n // 2].item()) == int(src[n // 2].item())
+    del src, dst
+    torch.cuda.empty_cache()
+
+
 def test_rccl_single_rank_bootstrap():
     lib = native.load()
     import ctypes
