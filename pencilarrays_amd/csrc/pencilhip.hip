@@ -438,7 +438,7 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
     }
 
     /* generic fallback */
-    {
+    if (esz == 16 || esz == 8 || esz == 4) {
         DescDev dd = to_dev(dn);
         const dim3 blocks(grid_for(dn.total, 256)); /* grid-stride kernel */
         if (esz == 16)
@@ -448,14 +448,32 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
             hipLaunchKernelGGL(k_copy_generic<uint64_t>, blocks,
                                dim3(256), 0, stream, (const uint64_t *)s,
                                (uint64_t *)d, dd);
-        else if (esz == 4)
+        else
             hipLaunchKernelGGL(k_copy_generic<uint32_t>, blocks,
                                dim3(256), 0, stream, (const uint32_t *)s,
                                (uint32_t *)d, dd);
-        else
-            return fail("unsupported element size %lld", (long long)esz);
         HIP_CHECK(hipGetLastError());
         return 0;
+    }
+
+    /* any other element size (the reference allows arbitrary isbits
+     * element types): byte-ify — prepend a stride-1 axis of esz bytes,
+     * which makes the descriptor linear-copyable at byte granularity
+     * (slow for tiny elements, always correct) */
+    {
+        if (dn.nd + 1 > MAXND) return fail("too many dims to byte-ify");
+        int64_t dims[MAXND + 1], ss[MAXND + 1], ds[MAXND + 1];
+        dims[0] = esz;
+        ss[0] = 1;
+        ds[0] = 1;
+        for (int a = 0; a < dn.nd; a++) {
+            dims[a + 1] = dn.dims[a];
+            ss[a + 1] = dn.sstr[a] * esz;
+            ds[a + 1] = dn.dstr[a] * esz;
+        }
+        CopyDescH b = normalize_desc(dn.nd + 1, dims, ss, dn.soff * esz, ds,
+                                     dn.doff * esz);
+        return launch_desc(b, 1, src, dst, stream);
     }
 }
 
