@@ -46,12 +46,14 @@ WORLD1 = [
     ((42, 31, 29), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.complex64),
     ((42, 31, 29), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.complex128),
     ((42, 31, 29), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.float32),
+    ((42, 31, 29), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.float16),
     ((24, 18, 12), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (3,), np.float64),
     ((24, 18, 12), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (4, 3), np.float64),
     ((1, 7, 5), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.float64),
 ]
 
 _T_DTYPE = {
+    np.dtype(np.float16): torch.float16,
     np.dtype(np.float64): torch.float64,
     np.dtype(np.float32): torch.float32,
     np.dtype(np.complex64): torch.complex64,
