@@ -22,12 +22,12 @@ either C or Fortran axis conventions in the storage itself.
 from __future__ import annotations
 
 import math
-from typing import Optional, Sequence, Tuple
+from typing import Optional, Tuple
 
 import numpy as np
 
 from .pencil import Pencil
-from .permutations import perm_apply, perm_inv
+from .permutations import perm_inv
 
 
 class PencilArray:
@@ -38,7 +38,6 @@ class PencilArray:
         self.extra_dims = tuple(int(e) for e in extra_dims)
         mem = tuple(pencil.size_local(rank, memory_order=True)) + self.extra_dims
         self.mem_dims = mem
-        n = math.prod(mem) if mem else 1
         if data_flat.ndim != 1 or data_flat.shape[0] != math.prod(mem):
             raise ValueError(
                 f"array has incorrect dimensions: {data_flat.shape}. "

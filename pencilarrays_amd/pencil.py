@@ -24,7 +24,6 @@ from .permutations import (
     Perm,
     check_perm,
     identity_perm,
-    is_identity,
     perm_apply,
 )
 from .topology import Topology

@@ -31,7 +31,7 @@ import json
 import math
 import os
 import sys
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 

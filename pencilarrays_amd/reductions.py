@@ -10,8 +10,7 @@ With no process group initialised (world = 1) the local value is global.
 
 from __future__ import annotations
 
-import math
-from typing import Callable, Optional, Union
+from typing import Callable, Optional
 
 import numpy as np
 

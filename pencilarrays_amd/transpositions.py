@@ -20,9 +20,7 @@ Execution backends:
 
 from __future__ import annotations
 
-import math
-import os
-from typing import List, Optional, Sequence
+from typing import Sequence
 
 import numpy as np
 
