@@ -509,6 +509,64 @@ int main()
         TRUNW("RS sweep 128x64 r16 c32", 128, 64, 16, 32);
         TRUNW("RS sweep 64x64 r8 c16", 64, 64, 8, 16);
         TRUNW("RS sweep 128x128 r16 c8", 128, 128, 16, 8);
+
+        /* within-probe interleaved A/B: rect (current) vs sweep c32/c64,
+         * 12 rounds each, report per-variant best and median */
+        {
+            const int ROUNDS = 12;
+            double t_rect_ms[ROUNDS], t_c32[ROUNDS], t_c64[ROUNDS];
+            hipEvent_t a, b;
+            CHK(hipEventCreate(&a));
+            CHK(hipEventCreate(&b));
+            const int64_t nti = NI / 128;
+            const int64_t ntj_t = NJ / 64;
+            const int64_t njc32 = (ntj_t + 31) / 32, njc64 = (ntj_t + 63) / 64;
+#define ONE(kern_launch, arr, r)                                             \
+    {                                                                        \
+        CHK(hipEventRecord(a));                                              \
+        kern_launch;                                                         \
+        CHK(hipEventRecord(b));                                              \
+        CHK(hipEventSynchronize(b));                                         \
+        float ms;                                                            \
+        CHK(hipEventElapsedTime(&ms, a, b));                                 \
+        arr[r] = ms;                                                         \
+    }
+            for (int r = 0; r < ROUNDS; r++) {
+                ONE(hipLaunchKernelGGL((t_rect<128, 64, 16>),
+                                       dim3((uint32_t)(nti * ntj_t)),
+                                       dim3(64, 16), 0, 0,
+                                       (const uint64_t *)s, (uint64_t *)d, NI,
+                                       NJ, nti),
+                    t_rect_ms, r);
+                ONE(hipLaunchKernelGGL((t_sweep<128, 64, 16, 32>),
+                                       dim3((uint32_t)(nti * njc32)),
+                                       dim3(64, 16), 0, 0,
+                                       (const uint64_t *)s, (uint64_t *)d, NI,
+                                       NJ, nti),
+                    t_c32, r);
+                ONE(hipLaunchKernelGGL((t_sweep<128, 64, 16, 64>),
+                                       dim3((uint32_t)(nti * njc64)),
+                                       dim3(64, 16), 0, 0,
+                                       (const uint64_t *)s, (uint64_t *)d, NI,
+                                       NJ, nti),
+                    t_c64, r);
+            }
+            CHK(hipEventDestroy(a));
+            CHK(hipEventDestroy(b));
+            const char *names[3] = {"rect(cur)", "sweep c32", "sweep c64"};
+            double *arrs[3] = {t_rect_ms, t_c32, t_c64};
+            for (int v = 0; v < 3; v++) {
+                double best = 1e30, sum = 0;
+                for (int r = 0; r < ROUNDS; r++) {
+                    if (arrs[v][r] < best) best = arrs[v][r];
+                    sum += arrs[v][r];
+                }
+                printf("AB %-10s best %8.1f GB/s  mean %8.1f GB/s\n",
+                       names[v], tio / (best * 1e-3) / 1e9,
+                       tio / ((sum / ROUNDS) * 1e-3) / 1e9);
+            }
+            fflush(stdout);
+        }
         /* t_sweep correctness */
         {
             const int64_t ni = 256, nj = 448; /* non-multiple of chunk span */
