@@ -159,24 +159,22 @@ __global__ __launch_bounds__(256) void k_copy_generic(const T *__restrict__ src,
  * is exactly this kernel; the reference's GPU path does a generic
  * permutedims! plus an extra temporary copy (:651-667 "TODO avoid
  * allocation") — here it is one kernel, no temporary. */
-template <typename T, int TILE_I, int TILE_J, int NROWS>
+template <typename T, int TILE_I, int TILE_J, int NROWS, int JCHUNK>
 __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
     const T *__restrict__ src, T *__restrict__ dst, DescDev d, int ta,
-    int64_t ntile_i, int64_t ntile_j, int64_t nblocks)
+    int64_t ntile_i, int64_t njchunk, int64_t nblocks)
 {
     __shared__ T tile[TILE_J][TILE_I + 1];
 
     const int tx = threadIdx.x; /* 0..63  : fast axis */
     const int ty = threadIdx.y; /* 0..NROWS-1 */
 
-    const int64_t tiles_per_batch = ntile_i * ntile_j;
-
-    int64_t bid = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t bid = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
     if (bid >= nblocks) return;
-    const int64_t tij = bid % tiles_per_batch;
-    int64_t batch = bid / tiles_per_batch;
-    const int64_t t_i = tij % ntile_i;
-    const int64_t t_j = tij / ntile_i;
+    const int64_t t_i = bid % ntile_i;
+    int64_t rest = bid / ntile_i;
+    const int64_t chunk = rest % njchunk;
+    int64_t batch = rest / njchunk;
 
     /* batch offsets over axes != 0, != ta */
     int64_t so_b = d.soff, do_b = d.doff;
@@ -188,29 +186,38 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
         do_b += j * d.dstr[a];
     }
 
-    const int64_t i0 = t_i * TILE_I;     /* along axis 0  */
-    const int64_t j0 = t_j * TILE_J;     /* along axis ta */
+    const int64_t i0 = t_i * TILE_I; /* along axis 0 */
     const int64_t ni = d.dims[0] - i0 < TILE_I ? d.dims[0] - i0 : TILE_I;
-    const int64_t nj = d.dims[ta] - j0 < TILE_J ? d.dims[ta] - j0 : TILE_J;
 
-    /* load: lanes sweep axis 0 (src-contiguous), rows sweep axis ta */
-    {
-        const int64_t base = so_b + i0 /* *1 */ + j0 * d.sstr[ta];
-        for (int j = ty; j < nj; j += NROWS) {
-            const int64_t row = base + (int64_t)j * d.sstr[ta];
-            for (int i = tx; i < ni; i += 64)
-                tile[j][i] = src[row + i];
+    /* sweep JCHUNK consecutive j-tiles with one workgroup: its reads walk
+     * one contiguous src region and each of its TILE_I output rows is
+     * written as a sequential stream (probe-measured +1.5% over one tile
+     * per WG at the real 1024^3-permuted shape, profiles/probe_ab). */
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = (chunk * JCHUNK + jt) * TILE_J; /* along ta */
+        if (j0 >= d.dims[ta]) break;
+        const int64_t nj = d.dims[ta] - j0 < TILE_J ? d.dims[ta] - j0 : TILE_J;
+
+        /* load: lanes sweep axis 0 (src-contiguous), rows sweep axis ta */
+        {
+            const int64_t base = so_b + i0 /* *1 */ + j0 * d.sstr[ta];
+            for (int j = ty; j < nj; j += NROWS) {
+                const int64_t row = base + (int64_t)j * d.sstr[ta];
+                for (int i = tx; i < ni; i += 64)
+                    tile[j][i] = src[row + i];
+            }
         }
-    }
-    __syncthreads();
-    /* store: lanes sweep axis ta (dst-contiguous), rows sweep axis 0 */
-    {
-        const int64_t base = do_b + j0 /* *1 */ + i0 * d.dstr[0];
-        for (int i = ty; i < ni; i += NROWS) {
-            const int64_t row = base + (int64_t)i * d.dstr[0];
-            for (int j = tx; j < nj; j += 64)
-                dst[row + j] = tile[j][i];
+        __syncthreads();
+        /* store: lanes sweep axis ta (dst-contiguous), rows sweep axis 0 */
+        {
+            const int64_t base = do_b + j0 /* *1 */ + i0 * d.dstr[0];
+            for (int i = ty; i < ni; i += NROWS) {
+                const int64_t row = base + (int64_t)i * d.dstr[0];
+                for (int j = tx; j < nj; j += 64)
+                    dst[row + j] = tile[j][i];
+            }
         }
+        __syncthreads(); /* before the next tile reuses the LDS */
     }
 }
 
@@ -399,39 +406,40 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
         int64_t nbatch = 1;
         for (int a = 1; a < dn.nd; a++)
             if (a != ta) nbatch *= dn.dims[a];
-        /* tile shapes probe-measured on MI355X (profiles/r01_probe_copy2):
-         * 8-B elements: 128(i)×64(j) r16 = 5453 GB/s (vs 64×64 r8 5269);
-         * 16-B: 32×32 (LDS-bounded).  Long reads, 512-B write bursts. */
+        /* tile shapes probe-measured on MI355X (profiles/r01_probe_copy*):
+         * 8/4-B elements: 128(i)x64(j) r16, sweeping 32 j-tiles per WG;
+         * 16-B: 32x32, one tile per WG (LDS-bounded). */
         if (esz == 8 || esz == 4) {
-            constexpr int TI = 128, TJ = 64, NR = 16;
+            constexpr int TI = 128, TJ = 64, NR = 16, JC = 32;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
-            const int64_t nblocks = nti * ntj * nbatch;
+            const int64_t njc = (ntj + JC - 1) / JC;
+            const int64_t nblocks = nti * njc * nbatch;
             dim3 grid;
             pa_status gst = grid2d(nblocks, 64 * NR, &grid);
             if (gst) return gst;
             if (esz == 8)
-                hipLaunchKernelGGL((k_transpose_tile<uint64_t, TI, TJ, NR>),
-                                   grid, dim3(64, NR), 0,
-                                   stream, (const uint64_t *)s, (uint64_t *)d,
-                                   dd, ta, nti, ntj, nblocks);
+                hipLaunchKernelGGL(
+                    (k_transpose_tile<uint64_t, TI, TJ, NR, JC>), grid,
+                    dim3(64, NR), 0, stream, (const uint64_t *)s,
+                    (uint64_t *)d, dd, ta, nti, njc, nblocks);
             else
-                hipLaunchKernelGGL((k_transpose_tile<uint32_t, TI, TJ, NR>),
-                                   grid, dim3(64, NR), 0,
-                                   stream, (const uint32_t *)s, (uint32_t *)d,
-                                   dd, ta, nti, ntj, nblocks);
+                hipLaunchKernelGGL(
+                    (k_transpose_tile<uint32_t, TI, TJ, NR, JC>), grid,
+                    dim3(64, NR), 0, stream, (const uint32_t *)s,
+                    (uint32_t *)d, dd, ta, nti, njc, nblocks);
         } else {
-            constexpr int TI = 32, TJ = 32, NR = 8;
+            constexpr int TI = 32, TJ = 32, NR = 8, JC = 1;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
             const int64_t nblocks = nti * ntj * nbatch;
             dim3 grid;
             pa_status gst = grid2d(nblocks, 64 * NR, &grid);
             if (gst) return gst;
-            hipLaunchKernelGGL((k_transpose_tile<uint4, TI, TJ, NR>),
-                               grid, dim3(64, NR), 0,
-                               stream, (const uint4 *)s, (uint4 *)d, dd, ta,
-                               nti, ntj, nblocks);
+            hipLaunchKernelGGL((k_transpose_tile<uint4, TI, TJ, NR, JC>),
+                               grid, dim3(64, NR), 0, stream,
+                               (const uint4 *)s, (uint4 *)d, dd, ta, nti, ntj,
+                               nblocks);
         }
         HIP_CHECK(hipGetLastError());
         return 0;
