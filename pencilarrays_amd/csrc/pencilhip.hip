@@ -410,24 +410,39 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
          * 8/4-B elements: 128(i)x64(j) r16, sweeping 32 j-tiles per WG;
          * 16-B: 32x32, one tile per WG (LDS-bounded). */
         if (esz == 8 || esz == 4) {
-            constexpr int TI = 128, TJ = 64, NR = 16, JC = 32;
+            constexpr int TI = 128, TJ = 64, NR = 16;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
-            const int64_t njc = (ntj + JC - 1) / JC;
+            /* adaptive sweep depth: long sweeps win on tall-skinny shapes
+             * (+1.5% A/B) but must keep >= ~4k workgroups for occupancy and
+             * tail balance on small/batched shapes */
+            int jc = 1;
+            if (ntj * nti * nbatch >= 32 * 4096 && ntj >= 256)
+                jc = 32;
+            else if (ntj * nti * nbatch >= 8 * 4096 && ntj >= 64)
+                jc = 8;
+            const int64_t njc = (ntj + jc - 1) / jc;
             const int64_t nblocks = nti * njc * nbatch;
             dim3 grid;
             pa_status gst = grid2d(nblocks, 64 * NR, &grid);
             if (gst) return gst;
-            if (esz == 8)
-                hipLaunchKernelGGL(
-                    (k_transpose_tile<uint64_t, TI, TJ, NR, JC>), grid,
-                    dim3(64, NR), 0, stream, (const uint64_t *)s,
-                    (uint64_t *)d, dd, ta, nti, njc, nblocks);
-            else
-                hipLaunchKernelGGL(
-                    (k_transpose_tile<uint32_t, TI, TJ, NR, JC>), grid,
-                    dim3(64, NR), 0, stream, (const uint32_t *)s,
-                    (uint32_t *)d, dd, ta, nti, njc, nblocks);
+#define LAUNCH_TT(T, JCV)                                                        hipLaunchKernelGGL((k_transpose_tile<T, TI, TJ, NR, JCV>), grid,                                dim3(64, NR), 0, stream, (const T *)s, (T *)d, dd,                           ta, nti, njc, nblocks)
+            if (esz == 8) {
+                if (jc == 32)
+                    LAUNCH_TT(uint64_t, 32);
+                else if (jc == 8)
+                    LAUNCH_TT(uint64_t, 8);
+                else
+                    LAUNCH_TT(uint64_t, 1);
+            } else {
+                if (jc == 32)
+                    LAUNCH_TT(uint32_t, 32);
+                else if (jc == 8)
+                    LAUNCH_TT(uint32_t, 8);
+                else
+                    LAUNCH_TT(uint32_t, 1);
+            }
+#undef LAUNCH_TT
         } else {
             constexpr int TI = 32, TJ = 32, NR = 8, JC = 1;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
