@@ -219,6 +219,86 @@ __global__ __launch_bounds__(64 * NROWS) void t_sweep(
     }
 }
 
+/* sweep + 16-B vector global accesses (interior tiles, TI=128: row = 64
+ * lanes x uint4) */
+template <int NROWS, int JCHUNK>
+__global__ __launch_bounds__(64 * NROWS) void t_sweep_vec(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    constexpr int TI = 128, TJ = 64;
+    __shared__ uint64_t tile[TJ][TI + 2];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TI;
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = (chunk * (int64_t)JCHUNK + jt) * TJ;
+        if (j0 >= NJ) break;
+        for (int j = ty; j < TJ; j += NROWS) {
+            const uint4 v = *(const uint4 *)&src[(i0 + 2 * tx) +
+                                                 NI * (j0 + j)];
+            tile[j][2 * tx] = ((const uint64_t *)&v)[0];
+            tile[j][2 * tx + 1] = ((const uint64_t *)&v)[1];
+        }
+        __syncthreads();
+        /* write: lanes 0..31 cover the 32 j-pairs of row i, lanes 32..63
+         * the j-pairs of row i+NROWS-offsetted partner (2 rows per y-step) */
+        {
+            const int jj = 2 * (tx & 31);
+            const int half = tx >> 5;
+            for (int i = 2 * ty + half; i < TI; i += 2 * NROWS) {
+                uint4 v;
+                ((uint64_t *)&v)[0] = tile[jj][i];
+                ((uint64_t *)&v)[1] = tile[jj + 1][i];
+                *(uint4 *)&dst[(j0 + jj) + NJ * (i0 + i)] = v;
+            }
+        }
+        __syncthreads();
+    }
+}
+
+/* sweep + double-buffered LDS (no barrier between store(jt) and load(jt+1));
+ * LDS 2x tiles => 1 WG/CU */
+template <int NROWS, int JCHUNK>
+__global__ __launch_bounds__(64 * NROWS) void t_sweep_db(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    constexpr int TI = 128, TJ = 64;
+    __shared__ uint64_t tile[2][TJ][TI + 1];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TI;
+    int cur = 0;
+    /* preload tile 0 */
+    {
+        const int64_t j0 = chunk * (int64_t)JCHUNK * TJ;
+        if (j0 >= NJ) return;
+        for (int j = ty; j < TJ; j += NROWS)
+            for (int i = tx; i < TI; i += 64)
+                tile[0][j][i] = src[(i0 + i) + NI * (j0 + j)];
+    }
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = (chunk * (int64_t)JCHUNK + jt) * TJ;
+        if (j0 >= NJ) break;
+        __syncthreads(); /* tile[cur] complete */
+        /* prefetch next tile into the other buffer while storing cur */
+        const int64_t j0n = j0 + TJ;
+        if (jt + 1 < JCHUNK && j0n < NJ)
+            for (int j = ty; j < TJ; j += NROWS)
+                for (int i = tx; i < TI; i += 64)
+                    tile[cur ^ 1][j][i] = src[(i0 + i) + NI * (j0n + j)];
+        for (int i = ty; i < TI; i += NROWS) {
+            const int64_t row = (j0) + NJ * (i0 + i);
+            for (int j = tx; j < TJ; j += 64)
+                dst[row + j] = tile[cur][j][i];
+        }
+        cur ^= 1;
+    }
+}
+
 /* vectorized 16-B loads/stores: interior tiles only (caller guarantees
  * NI,NJ multiples of TILE).  Lanes 0..31 load row 2*ty, lanes 32..63 row
  * 2*ty+1 (uint4 = 2 f64 along i).  Write phase symmetric along j. */
@@ -553,6 +633,45 @@ int main()
             }
             CHK(hipEventDestroy(a));
             CHK(hipEventDestroy(b));
+            /* extra variants, same best-of pattern */
+            {
+                const int64_t njc32 = (ntj_t + 31) / 32;
+                hipEvent_t a2, b2;
+                CHK(hipEventCreate(&a2));
+                CHK(hipEventCreate(&b2));
+#define BESTOF(name, launch)                                                 \
+    {                                                                        \
+        launch;                                                              \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < 8; r++) {                                        \
+            CHK(hipEventRecord(a2));                                         \
+            launch;                                                          \
+            CHK(hipEventRecord(b2));                                         \
+            CHK(hipEventSynchronize(b2));                                    \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a2, b2));                           \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("AB %-10s best %8.1f GB/s\n", name,                           \
+               tio / (best * 1e-3) / 1e9);                                   \
+        fflush(stdout);                                                      \
+    }
+                BESTOF("sweepvec32",
+                       hipLaunchKernelGGL((t_sweep_vec<16, 32>),
+                                          dim3((uint32_t)(nti * njc32)),
+                                          dim3(64, 16), 0, 0,
+                                          (const uint64_t *)s, (uint64_t *)d,
+                                          NI, NJ, nti));
+                BESTOF("sweepdb32",
+                       hipLaunchKernelGGL((t_sweep_db<16, 32>),
+                                          dim3((uint32_t)(nti * njc32)),
+                                          dim3(64, 16), 0, 0,
+                                          (const uint64_t *)s, (uint64_t *)d,
+                                          NI, NJ, nti));
+                CHK(hipEventDestroy(a2));
+                CHK(hipEventDestroy(b2));
+            }
             const char *names[3] = {"rect(cur)", "sweep c32", "sweep c64"};
             double *arrs[3] = {t_rect_ms, t_c32, t_c64};
             for (int v = 0; v < 3; v++) {
@@ -588,6 +707,36 @@ int main()
                     if (hd[j + nj * i] != hs[i + ni * j]) bad++;
             printf("t_sweep correctness: %s (%lld bad)\n", bad ? "FAIL" : "OK",
                    (long long)bad);
+            /* t_sweep_vec / t_sweep_db on the same pattern (ni,nj multiples
+             * of 128/64) */
+            {
+                const int64_t ni2 = 256, nj2 = 384;
+                CHK(hipMemset(d, 0xCC, ni2 * nj2 * 8));
+                const int64_t nti2 = ni2 / 128, njc2 = (nj2 / 64 + 3) / 4;
+                hipLaunchKernelGGL((t_sweep_vec<16, 4>),
+                                   dim3((uint32_t)(nti2 * njc2)),
+                                   dim3(64, 16), 0, 0, (const uint64_t *)s,
+                                   (uint64_t *)d, ni2, nj2, nti2);
+                CHK(hipMemcpy(hd, d, ni2 * nj2 * 8, hipMemcpyDeviceToHost));
+                int64_t bad2 = 0;
+                for (int64_t j = 0; j < nj2; j++)
+                    for (int64_t i = 0; i < ni2; i++)
+                        if (hd[j + nj2 * i] != hs[i + ni2 * j]) bad2++;
+                printf("t_sweep_vec correctness: %s (%lld bad)\n",
+                       bad2 ? "FAIL" : "OK", (long long)bad2);
+                CHK(hipMemset(d, 0xCC, ni2 * nj2 * 8));
+                hipLaunchKernelGGL((t_sweep_db<16, 4>),
+                                   dim3((uint32_t)(nti2 * njc2)),
+                                   dim3(64, 16), 0, 0, (const uint64_t *)s,
+                                   (uint64_t *)d, ni2, nj2, nti2);
+                CHK(hipMemcpy(hd, d, ni2 * nj2 * 8, hipMemcpyDeviceToHost));
+                bad2 = 0;
+                for (int64_t j = 0; j < nj2; j++)
+                    for (int64_t i = 0; i < ni2; i++)
+                        if (hd[j + nj2 * i] != hs[i + ni2 * j]) bad2++;
+                printf("t_sweep_db correctness: %s (%lld bad)\n",
+                       bad2 ? "FAIL" : "OK", (long long)bad2);
+            }
             free(hs);
             free(hd);
         }
