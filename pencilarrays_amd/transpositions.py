@@ -52,13 +52,26 @@ def _arrays_alias(a, b) -> bool:
 
 
 class Transposition:
-    def __init__(self, dest: PencilArray, src: PencilArray):
+    def __init__(self, dest: PencilArray, src: PencilArray,
+                 method: str = "grouped"):
+        """``Transposition(Ao, Ai; method)`` (Transpositions.jl:94-119).
+
+        ``method`` is accepted for signature parity with the reference's
+        PointToPoint/Alltoallv choice (:56-68): on MI355X both map to the
+        same grouped ncclSend/ncclRecv exchange over xGMI (RCCL has no
+        alltoallv; grouped p2p IS the alltoall realisation), so the value
+        ("grouped", "point_to_point", "alltoallv") does not change the
+        execution.
+        """
         if dest.extra_dims != src.extra_dims:
             raise ValueError(
                 f"incompatible number of extra dimensions of PencilArrays: "
                 f"{src.extra_dims} != {dest.extra_dims}")
         if dest.rank != src.rank:
             raise ValueError("dest and src must live on the same rank")
+        if method not in ("grouped", "point_to_point", "alltoallv"):
+            raise ValueError(f"unknown transpose method {method!r}")
+        self.method = method
         self.src = src
         self.dest = dest
         self.aliased = _arrays_alias(src.data, dest.data)
@@ -173,11 +186,12 @@ class Transposition:
         return self
 
 
-def transpose_into(dest: PencilArray, src: PencilArray) -> PencilArray:
-    """``transpose!(dest, src)`` (Transpositions.jl:161-169)."""
+def transpose_into(dest: PencilArray, src: PencilArray,
+                   method: str = "grouped") -> PencilArray:
+    """``transpose!(dest, src; method)`` (Transpositions.jl:161-169)."""
     if dest is src:
         return dest
-    return Transposition(dest, src).execute()
+    return Transposition(dest, src, method=method).execute()
 
 
 # ----------------------------------------------------------------------
