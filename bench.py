@@ -187,6 +187,20 @@ def main():
         dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
         elapsed = float(tmax.item())
 
+    # Self-verification (outside the timed region): the transpose is pure
+    # data movement, so the wrap-around sum of the int64 bit patterns is
+    # conserved exactly, whole-job, independent of order and distribution.
+    def bit_checksum(t):
+        v = t.view(torch.int64) if t.dtype != torch.int64 else t
+        s = v.sum()  # int64 wrap-around == modular arithmetic, exact
+        if dist:
+            dist.all_reduce(s, op=dist.ReduceOp.SUM)
+        return int(s.item())
+
+    cks_src = bit_checksum(src.data)
+    cks_dst = bit_checksum(dstz.data if t2 is not None else dst.data)
+    verify = "ok" if cks_src == cks_dst else "FAIL"
+
     if dist:
         dist.barrier()
         dist.destroy_process_group()
@@ -256,6 +270,7 @@ def main():
         "higher_is_better": True,
         "scaling": "strong",
         "vs_baseline": None,
+        "verify": verify,
         "dtype": "f64" if args.dtype == "float64" else "c64",
         "data": "synthetic",
         "config": {
