@@ -55,3 +55,21 @@ def test_fuzz_sim_vs_oracle():
         for r in range(nr):
             assert np.array_equal(back[r].data, parents[r]), \
                 f"trial {trial} rank {r} roundtrip"
+
+        # in-place variant of the same trial: one shared buffer per rank
+        pex = math.prod(extra) if extra else 1
+        bufs = []
+        for r in range(nr):
+            n = max(Pi.length_local(r), Po.length_local(r)) * pex
+            b = np.zeros(n, dtype=dtype)
+            b[:parents[r].size] = parents[r]
+            bufs.append(b)
+        srcs_ip = [PencilArray(Pi, r, bufs[r][:parents[r].size], extra)
+                   for r in range(nr)]
+        dests_ip = [PencilArray(Po, r,
+                                bufs[r][:Po.length_local(r) * pex], extra)
+                    for r in range(nr)]
+        run_transpose_sim(dests_ip, srcs_ip)
+        for r in range(nr):
+            assert np.array_equal(dests_ip[r].data, exp[r]), \
+                f"trial {trial} rank {r} in-place"
