@@ -52,6 +52,12 @@ SWEEP = [
     ((5, 4, 41), (4, 2), (0, 1), ID3, (2, 1), (2, 1, 0), (), np.complex128),
     # 2-D data
     ((17, 23), (2, 2), (0, 1), (0, 1), (0, 1), (1, 0), (), np.float64),
+    # 3-D (all-dims) decomposition, ComplexF32: only the permutation can
+    # change (pencils.jl:522-542)
+    ((16, 21, 41), (2, 2, 1), (0, 1, 2), ID3, (0, 1, 2), (1, 2, 0), (),
+     np.complex64),
+    ((12, 10, 8), (2, 1, 2), (0, 1, 2), (1, 2, 0), (0, 1, 2), (2, 1, 0), (),
+     np.complex64),
     # 4-D data (the reference is N-dimensional; Pencil{N,M} any N)
     ((6, 7, 8, 9), (2, 2), (1, 3), (0, 1, 2, 3), (0, 3), (3, 0, 1, 2), (),
      np.float64),
