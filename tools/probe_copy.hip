@@ -333,6 +333,73 @@ __global__ __launch_bounds__(64 * NROWS) void t_vec(
     }
 }
 
+/* ---------------- single-direction pattern ceilings ----------------
+ * Bound the transpose: what does the hardware give for its READ pattern
+ * alone (rows of TI contiguous f64) and its WRITE pattern alone (rows of
+ * TJ=64 contiguous f64 scattered at stride NJ), vs contiguous controls? */
+
+__global__ __launch_bounds__(1024) void r_pattern(const uint64_t *__restrict__ src,
+                                                  uint64_t *__restrict__ sink,
+                                                  int64_t NI, int64_t NJ,
+                                                  int64_t nti)
+{
+    /* mimic the sweep kernel's load phase: TI=128, TJ=64, NROWS=16, JC=32 */
+    const int tx = threadIdx.x & 63, ty = (threadIdx.x >> 6);
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * 128;
+    uint64_t acc = 0;
+    for (int jt = 0; jt < 32; jt++) {
+        const int64_t j0 = (chunk * 32 + jt) * 64;
+        if (j0 >= NJ) break;
+        for (int j = ty; j < 64; j += 16)
+            for (int i = tx; i < 128; i += 64)
+                acc ^= src[(i0 + i) + NI * (j0 + j)];
+    }
+    if (acc == 0xDEADBEEFCAFEBABEull) sink[0] = acc; /* keep loads live */
+}
+
+__global__ __launch_bounds__(1024) void w_pattern(uint64_t *__restrict__ dst,
+                                                  int64_t NI, int64_t NJ,
+                                                  int64_t nti)
+{
+    /* mimic the sweep kernel's store phase: rows i (TI=128), 64-elem bursts
+     * at stride NJ, sweeping 32 j-tiles */
+    const int tx = threadIdx.x & 63, ty = (threadIdx.x >> 6);
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * 128;
+    for (int jt = 0; jt < 32; jt++) {
+        const int64_t j0 = (chunk * 32 + jt) * 64;
+        if (j0 >= NJ) break;
+        for (int i = ty; i < 128; i += 16) {
+            const int64_t row = j0 + NJ * (i0 + i);
+            for (int j = tx; j < 64; j += 64)
+                dst[row + j] = (uint64_t)(row + j);
+        }
+    }
+}
+
+__global__ __launch_bounds__(256) void r_contig(const uint64_t *__restrict__ s,
+                                                uint64_t *__restrict__ sink,
+                                                int64_t n)
+{
+    const int64_t i = ((int64_t)blockIdx.y * gridDim.x + blockIdx.x) * 256 +
+                      threadIdx.x;
+    if (i < n) {
+        const uint64_t v = s[i];
+        if (v == 0xDEADBEEFCAFEBABEull) sink[0] = v;
+    }
+}
+
+__global__ __launch_bounds__(256) void w_contig(uint64_t *__restrict__ d,
+                                                int64_t n)
+{
+    const int64_t i = ((int64_t)blockIdx.y * gridDim.x + blockIdx.x) * 256 +
+                      threadIdx.x;
+    if (i < n) d[i] = (uint64_t)i;
+}
+
 /* ---------------- harness ---------------- */
 
 static double bench(void (*launch)(void *, void *, int64_t, int), void *s,
@@ -671,6 +738,60 @@ int main()
                                           NI, NJ, nti));
                 CHK(hipEventDestroy(a2));
                 CHK(hipEventDestroy(b2));
+            }
+            /* single-direction pattern ceilings (one-direction GB/s) */
+            {
+                const double one = 1.0 * NI * NJ * 8;
+                const int64_t njc32 = (ntj_t + 31) / 32;
+                hipEvent_t a3, b3;
+                CHK(hipEventCreate(&a3));
+                CHK(hipEventCreate(&b3));
+#define BEST1(name, launch, bytes)                                           \
+    {                                                                        \
+        launch;                                                              \
+        CHK(hipDeviceSynchronize());                                         \
+        double best = 1e30;                                                  \
+        for (int r = 0; r < 8; r++) {                                        \
+            CHK(hipEventRecord(a3));                                         \
+            launch;                                                          \
+            CHK(hipEventRecord(b3));                                         \
+            CHK(hipEventSynchronize(b3));                                    \
+            float ms;                                                        \
+            CHK(hipEventElapsedTime(&ms, a3, b3));                           \
+            if (ms < best) best = ms;                                        \
+        }                                                                    \
+        printf("DIR %-12s %8.1f GB/s (one direction)\n", name,               \
+               bytes / (best * 1e-3) / 1e9);                                 \
+        fflush(stdout);                                                      \
+    }
+                BEST1("read-pattern",
+                      hipLaunchKernelGGL(r_pattern,
+                                         dim3((uint32_t)(nti * njc32)),
+                                         dim3(1024), 0, 0,
+                                         (const uint64_t *)s, (uint64_t *)d,
+                                         NI, NJ, nti),
+                      one);
+                BEST1("write-pattern",
+                      hipLaunchKernelGGL(w_pattern,
+                                         dim3((uint32_t)(nti * njc32)),
+                                         dim3(1024), 0, 0, (uint64_t *)d, NI,
+                                         NJ, nti),
+                      one);
+                const int64_t n64 = NI * NJ;
+                BEST1("read-contig",
+                      hipLaunchKernelGGL(r_contig,
+                                         dim3((uint32_t)((n64 + 255) / 256)),
+                                         dim3(256), 0, 0,
+                                         (const uint64_t *)s, (uint64_t *)d,
+                                         n64),
+                      one);
+                BEST1("write-contig",
+                      hipLaunchKernelGGL(w_contig,
+                                         dim3((uint32_t)((n64 + 255) / 256)),
+                                         dim3(256), 0, 0, (uint64_t *)d, n64),
+                      one);
+                CHK(hipEventDestroy(a3));
+                CHK(hipEventDestroy(b3));
             }
             const char *names[3] = {"rect(cur)", "sweep c32", "sweep c64"};
             double *arrs[3] = {t_rect_ms, t_c32, t_c64};
