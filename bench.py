@@ -60,7 +60,7 @@ def parse_args():
     return ap.parse_args()
 
 
-def cpu_baseline(sample=(512, 512, 512), reps=3):
+def cpu_baseline(sample=(512, 512, 512), reps=8):
     """Time the line-faithful C oracle (the reference algorithm, OpenMP over
     this box's host cores) on a bounded sample of the same workload: world=1
     x->y, Float64 — kind 'port' (restatement), reported as context."""
