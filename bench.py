@@ -44,9 +44,16 @@ def parse_args():
     ap.add_argument("--size", type=int, nargs=3, default=[1024, 1024, 1024])
     ap.add_argument("--grid", type=int, nargs=2, default=None,
                     help="process grid P1 P2 (default: BASELINE grids)")
-    ap.add_argument("--permuted", action="store_true",
-                    help="config-5 variant: output pencil memory-permuted "
-                         "(1,2,0) — the PencilFFTs layout")
+    ap.add_argument("--permuted", action="store_true", default=True,
+                    help="output pencil memory-permuted (1,2,0) — the "
+                         "PencilFFTs layout (test/transpose.jl:29-30). "
+                         "DEFAULT: this is the realistic workload; at grid "
+                         "1x1 an identity-layout x->y collapses to a plain "
+                         "device copy")
+    ap.add_argument("--identity", dest="permuted", action="store_false",
+                    help="identity memory layout variant (at 1x1 this is a "
+                         "straight copy; reported as variant_identity in "
+                         "the default run)")
     ap.add_argument("--double", action="store_true",
                     help="config-4 variant: x->y->z chained double "
                          "transpose (PencilFFTs pattern), deferred waits")
@@ -103,7 +110,11 @@ def main():
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl", rank=rank, world_size=world)
+        # Control plane on GLOO: barriers/allreduce/uid-exchange only.  The
+        # engine's own RCCL communicators are then the ONLY NCCL-family
+        # comms per device (interleaving a torch-NCCL comm with the engine's
+        # grouped send/recv is a known deadlock hazard).
+        dist.init_process_group("gloo", rank=rank, world_size=world)
 
     torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
@@ -118,9 +129,6 @@ def main():
 
     topo = Topology(grid)
     Pi = Pencil(topo, dims, (1, 2))
-    po_perm = (1, 2, 0) if (args.permuted or args.double) else None
-    Po = Pencil(topo, dims, (0, 2), permute=po_perm)
-    Pz = Pencil(topo, dims, (0, 1), permute=(2, 1, 0)) if args.double else None
 
     gen = torch.Generator(device=device).manual_seed(0xC0FFEE + rank)
     n_in = Pi.length_local(rank)
@@ -133,78 +141,128 @@ def main():
         sl.copy_(torch.randn(sl.numel(), generator=gen, dtype=tdt,
                              device=device))
     src = PencilArray(Pi, rank, src_t)
-    dst = PencilArray(
-        Po, rank, torch.empty(Po.length_local(rank), dtype=tdt, device=device))
 
-    t = Transposition(dst, src)
-    t.execute()  # builds native plan, allocates staging, inits RCCL comms
-    t2 = None
-    if args.double:
-        dstz = PencilArray(
-            Pz, rank,
-            torch.empty(Pz.length_local(rank), dtype=tdt, device=device))
-        t2 = Transposition(dstz, dst)
-        t2.execute()
-    torch.cuda.synchronize()
+    def measure(permuted):
+        """One full measurement (setup + warmup + timed region + verify +
+        stage times) of the x->y (or x->y->z) transpose; returns a dict."""
+        po_perm = (1, 2, 0) if (permuted or args.double) else None
+        Po = Pencil(topo, dims, (0, 2), permute=po_perm)
+        Pz = (Pencil(topo, dims, (0, 1), permute=(2, 1, 0))
+              if args.double else None)
+        dst = PencilArray(Po, rank, torch.empty(
+            Po.length_local(rank), dtype=tdt, device=device))
 
-    stream = torch.cuda.current_stream()
+        t = Transposition(dst, src)
+        t.execute()  # builds native plan, binds staging, inits RCCL comms
+        t2 = dstz = None
+        if args.double:
+            dstz = PencilArray(Pz, rank, torch.empty(
+                Pz.length_local(rank), dtype=tdt, device=device))
+            t2 = Transposition(dstz, dst)
+            t2.execute()
+        torch.cuda.synchronize()
 
-    def step(sync=False):
-        # deferred waits (waitall=false pattern): both hops enqueue, the
-        # stream orders them
-        t._native.execute(src.data, dst.data, sync=False)
-        if t2 is not None:
-            t2._native.execute(dst.data, dstz.data, sync=False)
-        if sync:
+        stream = torch.cuda.current_stream()
+
+        def step(sync=False):
+            # deferred waits (waitall=false pattern): hops enqueue, the
+            # stream orders them
+            t._native.execute(src.data, dst.data, sync=False)
+            if t2 is not None:
+                t2._native.execute(dst.data, dstz.data, sync=False)
+            if sync:
+                torch.cuda.synchronize()
+
+        for _ in range(args.warmup):
+            step()
+        torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
-        step()
-    torch.cuda.synchronize()
-    if dist:
-        dist.barrier()
+        # HIP events around the timed region on the launch stream: per-step
+        # kernel-side duration for the roofline (single-kernel steps at N=1).
+        ev_a = torch.cuda.Event(enable_timing=True)
+        ev_b = torch.cuda.Event(enable_timing=True)
+        ev_a.record(stream)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        ev_b.record(stream)
         torch.cuda.synchronize()
-
-    # HIP events around the timed region on the launch stream: per-step
-    # kernel-side duration for the roofline (single-kernel steps at N=1).
-    ev_a = torch.cuda.Event(enable_timing=True)
-    ev_b = torch.cuda.Event(enable_timing=True)
-    ev_a.record(stream)
-    t0 = time.perf_counter()
-    for _ in range(args.steps):
-        step()
-    ev_b.record(stream)
-    torch.cuda.synchronize()
-    t1 = time.perf_counter()
-    if dist:
-        dist.barrier()
-        torch.cuda.synchronize()
-    elapsed = t1 - t0
-    gpu_ms = ev_a.elapsed_time(ev_b)
-
-    if dist:
-        tmax = torch.tensor([elapsed], device=device)
-        dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
-        elapsed = float(tmax.item())
-
-    # Self-verification (outside the timed region): the transpose is pure
-    # data movement, so the wrap-around sum of the int64 bit patterns is
-    # conserved exactly, whole-job, independent of order and distribution.
-    def bit_checksum(t):
-        v = t.view(torch.int64) if t.dtype != torch.int64 else t
-        s = v.sum()  # int64 wrap-around == modular arithmetic, exact
+        t1 = time.perf_counter()
         if dist:
-            dist.all_reduce(s, op=dist.ReduceOp.SUM)
-        return int(s.item())
+            dist.barrier()
+            torch.cuda.synchronize()
+        elapsed = t1 - t0
+        gpu_ms = ev_a.elapsed_time(ev_b)
 
-    cks_src = bit_checksum(src.data)
-    cks_dst = bit_checksum(dstz.data if t2 is not None else dst.data)
-    verify = "ok" if cks_src == cks_dst else "FAIL"
+        if dist:
+            tmax = torch.tensor([elapsed])  # CPU tensor: gloo
+            dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
+            elapsed = float(tmax.item())
+
+        # Per-stage HIP-event times (TimerOutputs analogue): one extra
+        # untimed step with timing events enabled, AFTER the timed region.
+        stage_ms = None
+        try:
+            t._native.native.enable_timing(True)
+            if t2 is not None:
+                t2._native.native.enable_timing(True)
+            step(sync=True)
+
+            def fmt(native):
+                return {k: (round(v, 4) if v is not None else None)
+                        for k, v in native.stage_times().items()}
+            stage_ms = fmt(t._native.native)
+            t._native.native.enable_timing(False)
+            if t2 is not None:
+                stage_ms = {"hop_xy": stage_ms,
+                            "hop_yz": fmt(t2._native.native)}
+                t2._native.native.enable_timing(False)
+        except Exception as exc:  # never fail the bench over telemetry
+            stage_ms = {"error": str(exc)}
+
+        # Self-verification (outside the timed region): the transpose is
+        # pure data movement, so the wrap-around sum of the int64 bit
+        # patterns is conserved exactly, whole-job, independent of order
+        # and distribution.
+        def bit_checksum(ten):
+            v = ten.view(torch.int64) if ten.dtype != torch.int64 else ten
+            s = v.sum().cpu()  # int64 wrap-around == modular, exact
+            if dist:
+                dist.all_reduce(s, op=dist.ReduceOp.SUM)
+            return int(s.item())
+
+        cks_src = bit_checksum(src.data)
+        cks_dst = bit_checksum(dstz.data if t2 is not None else dst.data)
+        return {
+            "elapsed": elapsed,
+            "gpu_ms": gpu_ms,
+            "verify": "ok" if cks_src == cks_dst else "FAIL",
+            "stage_ms": stage_ms,
+            "permuted": permuted,
+        }
+
+    m = measure(args.permuted)
+    # the identity-layout variant (at 1x1: a straight device copy) measured
+    # alongside the headline when it differs from it (N=1 default runs)
+    variant = None
+    if n_gpus == 1 and args.permuted and not args.double:
+        mi = measure(False)
+        variant = {
+            "workload": "identity memory layout (collapses to device copy)",
+            "value": round(math.prod(dims) * esz /
+                           (mi["elapsed"] / args.steps) / 2**30, 2),
+            "ms_per_step": round(mi["elapsed"] / args.steps * 1e3, 4),
+            "verify": mi["verify"],
+        }
 
     if dist:
         dist.barrier()
         dist.destroy_process_group()
 
+    elapsed, gpu_ms, verify = m["elapsed"], m["gpu_ms"], m["verify"]
     ms_per_step = elapsed / args.steps * 1e3
     global_bytes = math.prod(dims) * esz
     nhops = 2 if args.double else 1
@@ -228,6 +286,9 @@ def main():
         # correction) + WRITE_SIZE == algorithmic bytes, no re-reads.
         measured_traffic = (algo_bytes
                             if tuple(dims) == (1024, 1024, 1024) else None)
+        kname = ("k_transpose_tile (fused local permuted copy)"
+                 if args.permuted or args.double
+                 else "k_copy_1d (fused local copy, identity layout)")
         roofline = {
             "bound": "hbm",
             "achieved": round(algo_bytes / (kernel_ms * 1e-3) / 1e9, 1),
@@ -238,7 +299,7 @@ def main():
             "traffic_source": ("rocprofv3 --pmc FETCH_SIZE/WRITE_SIZE, "
                                "profiles/r01_pmc_traffic.md"
                                if measured_traffic else None),
-            "kernel": "fused local permuted copy (k_copy_1d / k_transpose_tile)",
+            "kernel": kname,
         }
     else:
         local_bytes = global_bytes // n_gpus
@@ -285,7 +346,10 @@ def main():
         },
         "roofline": roofline,
         "cpu_baseline": cb,
+        "stage_ms": m["stage_ms"],
     }
+    if variant is not None:
+        result["variant_identity"] = variant
     print(json.dumps(result), flush=True)
 
 

@@ -116,6 +116,15 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
 /* MPI.Waitall(t) (Transpositions.jl:128-131): wait for completion. */
 pa_status pa_transpose_wait(pa_plan *p, void *stream);
 
+/* ---- per-stage timing -------------------------------------------------- */
+/* The TimerOutputs.@timeit_debug analogue (Transpositions.jl:173-177,327,
+ * 337): opt-in HIP-event timing of the last execute's stages.  Call
+ * pa_plan_stage_times after pa_transpose_wait; out = {pack_ms, local_ms,
+ * exchange_ms, unpack_ms} (-1 for stages the plan does not run; exchange is
+ * timed on the engine's comm stream, overlapping local_ms by design). */
+pa_status pa_plan_enable_timing(pa_plan *p, int enable);
+pa_status pa_plan_stage_times(pa_plan *p, double out[4]);
+
 /* ---- plan introspection (host-side, for parity tests) ---------------- */
 int pa_plan_nproc_sub(const pa_plan *p);      /* P (1 => purely local)  */
 int pa_plan_r_dim(const pa_plan *p);          /* R, or -1 if same decomp */

@@ -630,6 +630,15 @@ struct pa_plan {
      * default 1 (single grouped exchange). */
     int chunks = 1;
     std::vector<hipEvent_t> ev_chunks;
+    /* per-stage timing (the TimerOutputs analogue, Transpositions.jl:
+     * 173-177,327,337): opt-in HIP events with timing around pack / local /
+     * exchange / unpack of the LAST execute.  tm[0..1] pack start/end and
+     * tm[2] local end on the caller's stream; tm[3..4] exchange start/end on
+     * the comm stream; tm[5..6] unpack start/end on the caller's stream. */
+    bool timing = false;
+    hipEvent_t tm[7] = {nullptr, nullptr, nullptr, nullptr,
+                        nullptr, nullptr, nullptr};
+    bool tm_valid[7] = {false, false, false, false, false, false, false};
 };
 
 static int64_t prod_extra(const pa_plan &pl)
@@ -1121,6 +1130,8 @@ void pa_plan_destroy(pa_plan *p)
     if (p->ev_pack) (void)hipEventDestroy(p->ev_pack);
     if (p->ev_comm) (void)hipEventDestroy(p->ev_comm);
     for (auto e : p->ev_chunks) (void)hipEventDestroy(e);
+    for (auto e : p->tm)
+        if (e) (void)hipEventDestroy(e);
     if (p->comm_stream) (void)hipStreamDestroy(p->comm_stream);
     delete p;
 }
@@ -1172,9 +1183,23 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
         p->own_bufs = true;
     }
 
+    if (p->timing) {
+        for (auto &e : p->tm)
+            if (!e) HIP_CHECK(hipEventCreate(&e));
+        for (auto &v : p->tm_valid) v = false;
+    }
+#define TM_REC(i, strm)                                                      \
+    do {                                                                     \
+        if (p->timing) {                                                     \
+            HIP_CHECK(hipEventRecord(p->tm[i], strm));                       \
+            p->tm_valid[i] = true;                                           \
+        }                                                                    \
+    } while (0)
+
     /* 1. pack every remote block (Transpositions.jl:346-431); in aliased
      * mode also stage the self block into the recv tail before any dst
      * write (:394-404) */
+    TM_REC(0, stream);
     for (auto &blk : p->peers)
         if (blk.has_pack) {
             pa_status st =
@@ -1186,6 +1211,7 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
                                    p->recv_buf, stream);
         if (st) return st;
     }
+    TM_REC(1, stream);
 
     /* 2. exchange: grouped ncclSend/ncclRecv over xGMI on a dedicated comm
      * stream ordered after the pack kernels (replaces :419-428/:463-479;
@@ -1219,6 +1245,7 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
         }
         HIP_CHECK(hipEventRecord(p->ev_pack, stream));
         HIP_CHECK(hipStreamWaitEvent(p->comm_stream, p->ev_pack, 0));
+        TM_REC(3, p->comm_stream);
         /* C == 1: one grouped exchange.  C > 1: the exchange is split into C
          * groups of matching per-peer sub-blocks (outer rows of the block,
          * identical split on sender and receiver), and each chunk's unpack
@@ -1254,6 +1281,7 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
             NCCL_CHECK(ncclGroupEnd());
             HIP_CHECK(hipEventRecord(p->ev_chunks[c], p->comm_stream));
         }
+        TM_REC(4, p->comm_stream);
         HIP_CHECK(hipEventRecord(p->ev_comm, p->comm_stream));
     }
 
@@ -1269,6 +1297,7 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
                                    dst_parent, stream);
         if (st) return st;
     }
+    TM_REC(2, stream);
 
     /* 4. unpack received blocks (:489-536).  C == 1: all after the single
      * exchange completes.  C > 1: chunk c unpacks as soon as its group
@@ -1276,15 +1305,18 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
     if (C <= 1) {
         if (exchanging)
             HIP_CHECK(hipStreamWaitEvent(stream, p->ev_comm, 0));
+        TM_REC(5, stream);
         for (auto &blk : p->peers)
             if (blk.has_unpack) {
                 pa_status st = launch_desc(blk.unpack, p->esz, p->recv_buf,
                                            dst_parent, stream);
                 if (st) return st;
             }
+        TM_REC(6, stream);
     } else {
         for (int c = 0; c < C; c++) {
             HIP_CHECK(hipStreamWaitEvent(stream, p->ev_chunks[c], 0));
+            if (c == 0) TM_REC(5, stream);
             for (auto &blk : p->peers) {
                 if (!blk.has_unpack_raw) continue;
                 const int64_t lo = chunk_lo(blk.recv_outer, c, C);
@@ -1296,8 +1328,38 @@ pa_status pa_transpose_execute(pa_plan *p, const void *src_parent,
                 if (st) return st;
             }
         }
+        TM_REC(6, stream);
     }
+#undef TM_REC
 
+    return 0;
+}
+
+/* ---- per-stage timing (TimerOutputs analogue) ---------------------- */
+
+pa_status pa_plan_enable_timing(pa_plan *p, int enable)
+{
+    p->timing = enable != 0;
+    return 0;
+}
+
+pa_status pa_plan_stage_times(pa_plan *p, double out[4])
+{
+    /* Valid after the last execute has completed (pa_transpose_wait).
+     * out = {pack_ms, local_ms, exchange_ms, unpack_ms}; -1 for stages the
+     * plan does not have. */
+    if (!p->timing) return fail("timing not enabled (pa_plan_enable_timing)");
+    float ms;
+    auto span = [&](int a, int b) -> double {
+        if (!p->tm_valid[a] || !p->tm_valid[b]) return -1.0;
+        if (hipEventElapsedTime(&ms, p->tm[a], p->tm[b]) != hipSuccess)
+            return -1.0;
+        return (double)ms;
+    };
+    out[0] = span(0, 1); /* pack (+ staged self pack)      */
+    out[1] = span(1, 2); /* fused local / self unpack      */
+    out[2] = span(3, 4); /* exchange (comm stream)         */
+    out[3] = span(5, 6); /* unpack of received blocks      */
     return 0;
 }
 
