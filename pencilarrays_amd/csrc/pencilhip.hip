@@ -221,6 +221,78 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile(
     }
 }
 
+/* LDS-tiled transpose with VECTOR (16-B) STORES: 64(i) x 128(j) tile, each
+ * lane storing a j-pair as one uint4 — 1024-B write bursts per instruction.
+ * Interleaved-median probe (profiles/r2_probe_kernels*.txt): +1.5-1.8% over
+ * the scalar 128x64 sweep at the 1024^3-permuted shape (write path is
+ * L2-writeback-bound, wider bursts per instruction help where wider tiles
+ * do not).  8-byte elements only; caller guarantees every dst stride
+ * (except the unit ta axis) and the dst offset are EVEN and the dst base
+ * pointer is 16-B aligned, so every vector store is aligned.  Tail j-tiles
+ * (nj < TJ or odd) take a scalar epilogue. */
+template <typename T, int TI, int TJ, int NROWS, int JCHUNK>
+__global__ __launch_bounds__(64 * NROWS) void k_transpose_tile_vs(
+    const T *__restrict__ src, T *__restrict__ dst, DescDev d, int ta,
+    int64_t ntile_i, int64_t njchunk, int64_t nblocks)
+{
+    static_assert(sizeof(T) == 8, "vector-store tile is 8-byte-elem only");
+    __shared__ T tile[TJ][TI + 2];
+
+    const int tx = threadIdx.x;
+    const int ty = threadIdx.y;
+
+    const int64_t bid = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    if (bid >= nblocks) return;
+    const int64_t t_i = bid % ntile_i;
+    int64_t rest = bid / ntile_i;
+    const int64_t chunk = rest % njchunk;
+    int64_t batch = rest / njchunk;
+
+    int64_t so_b = d.soff, do_b = d.doff;
+    for (int a = 1; a < d.nd; a++) {
+        if (a == ta) continue;
+        const int64_t j = batch % d.dims[a];
+        batch /= d.dims[a];
+        so_b += j * d.sstr[a];
+        do_b += j * d.dstr[a];
+    }
+
+    const int64_t i0 = t_i * TI;
+    const int64_t ni = d.dims[0] - i0 < TI ? d.dims[0] - i0 : TI;
+
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = (chunk * JCHUNK + jt) * TJ;
+        if (j0 >= d.dims[ta]) break;
+        const int64_t nj = d.dims[ta] - j0 < TJ ? d.dims[ta] - j0 : TJ;
+
+        {
+            const int64_t base = so_b + i0 + j0 * d.sstr[ta];
+            for (int j = ty; j < nj; j += NROWS) {
+                const int64_t row = base + (int64_t)j * d.sstr[ta];
+                for (int i = tx; i < ni; i += 64)
+                    tile[j][i] = src[row + i];
+            }
+        }
+        __syncthreads();
+        {
+            const int64_t base = do_b + j0 + i0 * d.dstr[0];
+            const int64_t njv = nj & ~(int64_t)1; /* even part */
+            for (int i = ty; i < ni; i += NROWS) {
+                T *row = dst + base + (int64_t)i * d.dstr[0];
+                for (int j2 = 2 * tx; j2 < njv; j2 += 128) {
+                    uint4 q;
+                    ((T *)&q)[0] = tile[j2][i];
+                    ((T *)&q)[1] = tile[j2 + 1][i];
+                    *(uint4 *)&row[j2] = q;
+                }
+                if ((nj & 1) && tx == 0) /* odd tail element */
+                    row[nj - 1] = tile[nj - 1][i];
+            }
+        }
+        __syncthreads();
+    }
+}
+
 /* ================= descriptor normalization & dispatch ============= */
 
 struct CopyDescH {
@@ -406,9 +478,49 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
         int64_t nbatch = 1;
         for (int a = 1; a < dn.nd; a++)
             if (a != ta) nbatch *= dn.dims[a];
-        /* tile shapes probe-measured on MI355X (profiles/r01_probe_copy*):
-         * 8/4-B elements: 128(i)x64(j) r16, sweeping 32 j-tiles per WG;
-         * 16-B: 32x32, one tile per WG (LDS-bounded). */
+        /* tile shapes probe-measured on MI355X (profiles/r01_probe_copy*,
+         * r2_probe_kernels*): 8-B elements prefer the 64x128 vector-store
+         * tile (+1.5-1.8% median) when alignment permits; otherwise (and
+         * for 4-B) the scalar 128(i)x64(j) r16 sweep; 16-B: 32x32. */
+        bool vec_ok = false;
+        if (esz == 8) {
+            vec_ok = ((dn.doff & 1) == 0) && ((dn.dstr[0] & 1) == 0) &&
+                     (((uintptr_t)d & 15) == 0);
+            for (int a2 = 1; a2 < dn.nd && vec_ok; a2++)
+                if (a2 != ta && (dn.dstr[a2] & 1)) vec_ok = false;
+        }
+        if (vec_ok) {
+            constexpr int TI = 64, TJ = 128, NR = 16;
+            const int64_t nti = (dn.dims[0] + TI - 1) / TI;
+            const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
+            int64_t nbatch = 1;
+            for (int a2 = 1; a2 < dn.nd; a2++)
+                if (a2 != ta) nbatch *= dn.dims[a2];
+            int jc = 1; /* same occupancy/tail-balance thresholds as below */
+            if (ntj * nti * nbatch >= 32 * 4096 && ntj >= 256)
+                jc = 32;
+            else if (ntj * nti * nbatch >= 8 * 4096 && ntj >= 64)
+                jc = 8;
+            const int64_t njc = (ntj + jc - 1) / jc;
+            const int64_t nblocks = nti * njc * nbatch;
+            DescDev dd = to_dev(dn);
+            dim3 grid;
+            pa_status gst = grid2d(nblocks, 64 * NR, &grid);
+            if (gst) return gst;
+#define LAUNCH_VS(JCV)                                                       \
+    hipLaunchKernelGGL((k_transpose_tile_vs<uint64_t, TI, TJ, NR, JCV>),     \
+                       grid, dim3(64, NR), 0, stream, (const uint64_t *)s,   \
+                       (uint64_t *)d, dd, ta, nti, njc, nblocks)
+            if (jc == 32)
+                LAUNCH_VS(32);
+            else if (jc == 8)
+                LAUNCH_VS(8);
+            else
+                LAUNCH_VS(1);
+#undef LAUNCH_VS
+            HIP_CHECK(hipGetLastError());
+            return 0;
+        }
         if (esz == 8 || esz == 4) {
             constexpr int TI = 128, TJ = 64, NR = 16;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
