@@ -108,6 +108,18 @@ class Transposition:
         import torch
         import torch.distributed as dist
 
+        # one process per rank: dist rank/world must match the topology
+        # (a mismatch would silently address the wrong peers)
+        topo = plan.Pi.topology
+        if dist.get_world_size() != topo.nranks:
+            raise RuntimeError(
+                f"torch.distributed world size {dist.get_world_size()} != "
+                f"topology nranks {topo.nranks}")
+        if dist.get_rank() != plan.rank:
+            raise RuntimeError(
+                f"torch.distributed rank {dist.get_rank()} != topology "
+                f"rank {plan.rank}")
+
         send_buf = np.empty(plan.send_nelem_total, dtype=src_flat.dtype)
         recv_buf = np.empty(plan.recv_nelem_total, dtype=src_flat.dtype)
 

@@ -146,3 +146,30 @@ def test_read_type_mismatch_raises(tmp_path):
     bad = PencilArray.empty(pen, 0, dtype="float32")
     with pytest.raises(TypeError):
         MPIIOFile(path, "r", rank=0).read("u", bad)
+
+
+def test_read_size_bytes_mismatch_raises(tmp_path):
+    """check_metadata's size_bytes assert (mpi_io.jl:306-307): corrupted
+    metadata must be rejected, not read past."""
+    dims = (8, 6, 5)
+    g, pen, xs = _arrays(dims, (1, 1), (1, 2), (0, 1, 2), (), np.float64)
+    path = str(tmp_path / "t.bin")
+    f = MPIIOFile(path, "w", rank=0)
+    f.write("u", xs[0])
+    f.close()
+    meta = json.load(open(path + ".json"))
+    meta["datasets"]["u"]["size_bytes"] += 8
+    json.dump(meta, open(path + ".json", "w"))
+    y = PencilArray.empty(pen, 0)
+    with pytest.raises(ValueError, match="size_bytes"):
+        MPIIOFile(path, "r", rank=0).read("u", y)
+
+
+def test_write_unsupported_dtype_clear_error(tmp_path):
+    """float16 now maps to Float16; a truly unmapped dtype gets a clear
+    TypeError, not a KeyError."""
+    from pencilarrays_amd.pencilio import _julia_type
+    assert _julia_type(np.dtype(np.float16)) == "Float16"
+    assert _julia_type(np.dtype(np.int8)) == "Int8"
+    with pytest.raises(TypeError, match="unsupported element type"):
+        _julia_type(np.dtype([("a", np.int32)]))  # structured dtype
