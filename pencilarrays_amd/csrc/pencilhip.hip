@@ -104,6 +104,23 @@ __global__ __launch_bounds__(256) void k_copy_1d(const T *__restrict__ src,
     if (i < n) dst[i] = src[i];
 }
 
+/* Nontemporal variant for streams far past the 256 MiB Infinity Cache:
+ * bypassing L2/LLC gains ~5% at >=8 GiB payloads but LOSES ~7% when the
+ * working set is cache-resident (probe: P1D direct/nt-direct at 128 MiB /
+ * 8 GiB / 32 GiB, profiles/r2_probe_kernels8/9.txt) — dispatch gates on
+ * payload size. */
+typedef unsigned int v4u_nt __attribute__((ext_vector_type(4)));
+__global__ __launch_bounds__(256) void k_copy_1d_nt(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n) {
+        v4u_nt v = __builtin_nontemporal_load((const v4u_nt *)&src[i]);
+        __builtin_nontemporal_store(v, (v4u_nt *)&dst[i]);
+    }
+}
+
 /* Batched linear runs: axis 0 contiguous on both sides (coalesced); outer
  * axes strided.  Used for pack (strided window -> contiguous buffer when the
  * fastest axis survives) and its inverse.  Direct mapping (see k_copy_1d). */
@@ -423,7 +440,11 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
             dim3 blocks;
             pa_status gst = grid2d(grid_exact(w.total, 256), 256, &blocks);
             if (gst) return gst;
-            if (W == 16)
+            if (W == 16 && w.total * 16 >= (512ll << 20))
+                hipLaunchKernelGGL(k_copy_1d_nt, blocks, dim3(256),
+                                   0, stream, (const uint4 *)s + w.soff,
+                                   (uint4 *)d + w.doff, w.total);
+            else if (W == 16)
                 hipLaunchKernelGGL(k_copy_1d<uint4>, blocks, dim3(256),
                                    0, stream, (const uint4 *)s + w.soff,
                                    (uint4 *)d + w.doff, w.total);

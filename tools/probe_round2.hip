@@ -164,6 +164,36 @@ __global__ __launch_bounds__(64 * NROWS) void t_sweep_ntall(
     }
 }
 
+/* ---- giant 1-D copy kernels ---- */
+
+__global__ __launch_bounds__(256) void g_copy_direct(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n) dst[i] = src[i];
+}
+
+__global__ __launch_bounds__(256) void g_copy_gs(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = src[i];
+}
+
+typedef unsigned int v4ug __attribute__((ext_vector_type(4)));
+__global__ __launch_bounds__(256) void g_copy_direct_nt(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n) {
+        v4ug v = __builtin_nontemporal_load((const v4ug *)&src[i]);
+        __builtin_nontemporal_store(v, (v4ug *)&dst[i]);
+    }
+}
+
 /* ---- parameterized single-direction ceilings ---- */
 
 template <int TI, int NROWS, int JCHUNK>
@@ -431,6 +461,152 @@ int main()
                    tio / (best * 1e-3) / 1e9, tio / (med * 1e-3) / 1e9);
         }
         fflush(stdout);
+    }
+
+    /* TRUE headline shape: NI=1024, NJ=1048576 (8 GiB x2) — jc sweep for
+     * the vs kernel at the exact 1024^3 descriptor (the 262144 probe shape
+     * under-predicted the real gain; pick jc on the real shape). */
+    {
+        const int64_t NIr = 1024, NJr = 1048576;
+        const int64_t rbytes = NIr * NJr * 8;
+        void *s2, *d2;
+        CHK(hipMalloc(&s2, rbytes));
+        CHK(hipMalloc(&d2, rbytes));
+        CHK(hipMemset(s2, 0xA5, rbytes));
+        const double rio = 2.0 * rbytes;
+#define RSWEEP(name, kern, TI, TJ, NR, JC)                                   \
+    {                                                                        \
+        const int64_t nti = (NIr + TI - 1) / TI;                             \
+        const int64_t ntj = (NJr + TJ - 1) / TJ;                             \
+        const int64_t njc = (ntj + JC - 1) / JC;                             \
+        BEST(name,                                                           \
+             hipLaunchKernelGGL((kern<TI, TJ, NR, JC>),                      \
+                                dim3((uint32_t)(nti * njc)), dim3(64, NR),   \
+                                0, 0, (const uint64_t *)s2, (uint64_t *)d2,  \
+                                NIr, NJr, nti),                              \
+             rio);                                                           \
+    }
+        RSWEEP("TRUE vs 64x128 c8", t_sweep_vs, 64, 128, 16, 8);
+        RSWEEP("TRUE vs 64x128 c16", t_sweep_vs, 64, 128, 16, 16);
+        RSWEEP("TRUE vs 64x128 c32 (cur)", t_sweep_vs, 64, 128, 16, 32);
+        RSWEEP("TRUE vs 64x128 c64", t_sweep_vs, 64, 128, 16, 64);
+        RSWEEP("TRUE vs 64x128 c128", t_sweep_vs, 64, 128, 16, 128);
+        RSWEEP("TRUE scalar 128x64 c32", t_sweep_c, 128, 64, 16, 32);
+        RSWEEP("TRUE vsnt 64x128 c32", t_sweep_vsnt, 64, 128, 16, 32);
+        /* decisive interleaved A/B at THE shape: vs vs vsnt, 16 rounds */
+        {
+            const int ROUNDS = 16;
+            const int64_t nti = (NIr + 63) / 64;
+            const int64_t njc = (NJr / 128 + 31) / 32;
+            double mV[ROUNDS], mN[ROUNDS];
+            for (int r = 0; r < ROUNDS; r++) {
+                ONE2(hipLaunchKernelGGL((t_sweep_vs<64, 128, 16, 32>),
+                                        dim3((uint32_t)(nti * njc)),
+                                        dim3(64, 16), 0, 0,
+                                        (const uint64_t *)s2, (uint64_t *)d2,
+                                        NIr, NJr, nti),
+                     mV, r);
+                ONE2(hipLaunchKernelGGL((t_sweep_vsnt<64, 128, 16, 32>),
+                                        dim3((uint32_t)(nti * njc)),
+                                        dim3(64, 16), 0, 0,
+                                        (const uint64_t *)s2, (uint64_t *)d2,
+                                        NIr, NJr, nti),
+                     mN, r);
+            }
+            double *ar2[2] = {mV, mN};
+            const char *nm2[2] = {"vs  c32", "vsnt c32"};
+            for (int v = 0; v < 2; v++) {
+                double best = 1e30;
+                for (int r = 0; r < ROUNDS; r++)
+                    if (ar2[v][r] < best) best = ar2[v][r];
+                for (int i = 1; i < ROUNDS; i++) {
+                    double x = ar2[v][i];
+                    int j = i - 1;
+                    while (j >= 0 && ar2[v][j] > x) {
+                        ar2[v][j + 1] = ar2[v][j];
+                        j--;
+                    }
+                    ar2[v][j + 1] = x;
+                }
+                double med =
+                    0.5 * (ar2[v][ROUNDS / 2 - 1] + ar2[v][ROUNDS / 2]);
+                printf("TRUEAB %-8s best %8.1f  median %8.1f GB/s\n", nm2[v],
+                       rio / (best * 1e-3) / 1e9, rio / (med * 1e-3) / 1e9);
+            }
+            fflush(stdout);
+        }
+        CHK(hipFree(s2));
+        CHK(hipFree(d2));
+    }
+
+    /* 1-D copy payload sweep: NT-direct was never probed in r01 (only
+     * NT+grid-stride).  128 MiB (cache-resident), 8 GiB (1024^3 identity),
+     * 32 GiB (2048^3 identity). */
+    for (int pay = 0; pay < 3; pay++) {
+        const int64_t gib[3] = {(128LL << 20), (8LL << 30), (32LL << 30)};
+        const int64_t n16p = gib[pay] / 16;
+        void *sp, *dp;
+        CHK(hipMalloc(&sp, n16p * 16));
+        CHK(hipMalloc(&dp, n16p * 16));
+        CHK(hipMemset(sp, 0x3C, n16p * 16));
+        const double pio = 2.0 * n16p * 16;
+        auto directp = [&](int64_t n) {
+            const int64_t blocks = (n + 255) / 256;
+            const int64_t gx = blocks < 16777215 ? blocks : 16777215;
+            const int64_t gy = (blocks + gx - 1) / gx;
+            return dim3((uint32_t)gx, (uint32_t)gy);
+        };
+        char nm1[64], nm2[64];
+        snprintf(nm1, sizeof nm1, "P1D direct %lld MiB",
+                 (long long)(gib[pay] >> 20));
+        snprintf(nm2, sizeof nm2, "P1D nt-direct %lld MiB",
+                 (long long)(gib[pay] >> 20));
+        BEST(nm1,
+             hipLaunchKernelGGL(g_copy_direct, directp(n16p), dim3(256), 0,
+                                0, (const uint4 *)sp, (uint4 *)dp, n16p),
+             pio);
+        BEST(nm2,
+             hipLaunchKernelGGL(g_copy_direct_nt, directp(n16p), dim3(256),
+                                0, 0, (const uint4 *)sp, (uint4 *)dp, n16p),
+             pio);
+        CHK(hipFree(sp));
+        CHK(hipFree(dp));
+    }
+
+    /* giant 1-D copy (the 2048^3 identity path: 32 GiB each way — the
+     * direct grid was picked on 4 GiB payloads; re-probe at scale) */
+    {
+        const int64_t n16g = (32LL << 30) / 16; /* 32 GiB of uint4 */
+        void *sg, *dg;
+        CHK(hipMalloc(&sg, n16g * 16));
+        CHK(hipMalloc(&dg, n16g * 16));
+        CHK(hipMemset(sg, 0x3C, n16g * 16));
+        const double gio = 2.0 * n16g * 16;
+        /* direct: one elem/thread, 2-D grid like the production kernel */
+        auto direct = [&](int64_t n) {
+            const int64_t blocks = (n + 255) / 256;
+            const int64_t gx = blocks < 16777215 ? blocks : 16777215;
+            const int64_t gy = (blocks + gx - 1) / gx;
+            return dim3((uint32_t)gx, (uint32_t)gy);
+        };
+        BEST("G1D direct (cur)",
+             hipLaunchKernelGGL(g_copy_direct, direct(n16g), dim3(256), 0,
+                                0, (const uint4 *)sg, (uint4 *)dg, n16g),
+             gio);
+        BEST("G1D gridstride 8192",
+             hipLaunchKernelGGL(g_copy_gs, dim3(8192), dim3(256), 0, 0,
+                                (const uint4 *)sg, (uint4 *)dg, n16g),
+             gio);
+        BEST("G1D gridstride 32768",
+             hipLaunchKernelGGL(g_copy_gs, dim3(32768), dim3(256), 0, 0,
+                                (const uint4 *)sg, (uint4 *)dg, n16g),
+             gio);
+        BEST("G1D nt direct",
+             hipLaunchKernelGGL(g_copy_direct_nt, direct(n16g), dim3(256),
+                                0, 0, (const uint4 *)sg, (uint4 *)dg, n16g),
+             gio);
+        CHK(hipFree(sg));
+        CHK(hipFree(dg));
     }
 
     /* cache-resident small shape (the 256^3 class: 128 MiB payload fits the
