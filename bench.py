@@ -29,6 +29,11 @@ import time
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
+# the host driver only supports dmabuf IPC; without this, RCCL across
+# processes fails with hipIpcGetMemHandle errors (normally exported by the
+# environment — kept here as insurance for the multi-process run)
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
 import torch  # noqa: E402
 
 from pencilarrays_amd import Pencil, PencilArray, Topology, Transposition  # noqa: E402
