@@ -302,7 +302,7 @@ def main():
             "frac": round(algo_bytes / (kernel_ms * 1e-3) / 8e12, 4),
             "traffic": measured_traffic,
             "traffic_source": ("rocprofv3 --pmc FETCH_SIZE/WRITE_SIZE, "
-                               "profiles/r01_pmc_traffic.md"
+                               "profiles/r2_kernel_stats.md"
                                if measured_traffic else None),
             "kernel": kname,
         }
