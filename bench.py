@@ -247,6 +247,9 @@ def main():
             "verify": "ok" if cks_src == cks_dst else "FAIL",
             "stage_ms": stage_ms,
             "permuted": permuted,
+            # real subgroup size of the transposed dimension (the exchange
+            # group), from the plan — not assumed from the grid shape
+            "nproc_sub": t._native.native.nproc_sub,
         }
 
     m = measure(args.permuted)
@@ -279,9 +282,10 @@ def main():
     # Roofline of the dominant kernel.  At N=1 the whole step is the fused
     # local copy: algorithmic HBM traffic = 2 * elem_size per global element
     # (read + write; the reference's staged path moves 4x).  At
-    # N>1 the dominant resource is the xGMI link: bytes crossing one link
-    # per step = remote fraction of the local block, both directions.
-    P_sub = grid[0]
+    # N>1 the dominant resource is the xGMI link: each of the P_sub-1 peers
+    # rides its own point-to-point link, so per-link bytes per step =
+    # local_bytes/P_sub each direction (the remote fraction split evenly).
+    P_sub = m["nproc_sub"]
     hops = nhops
     if n_gpus == 1:
         kernel_ms = gpu_ms / args.steps / hops
