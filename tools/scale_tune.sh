@@ -31,6 +31,6 @@ echo "=== chunking (default grid) ==="
 run --chunks 2
 run --chunks 4
 run --chunks 8
-echo "=== permuted + double at the best grid (re-run after reading above) ==="
-run --permuted
+echo "=== identity + double at the best grid (permuted is the default) ==="
+run --identity
 run --double
