@@ -327,3 +327,51 @@ def test_native_allreduce_tensor_world1():
             native.allreduce_tensor(comm, t, op)
             torch.cuda.synchronize()
             assert torch.equal(t, before), (dt, op)
+
+
+def test_pool_growth_with_async_work_in_flight():
+    """Pool growth while deferred-wait executes are still enqueued: the
+    reserve() device-sync must drain the in-flight chain before retiring
+    the old buffers; afterwards the old plan re-binds and stays exact."""
+    topo = Topology((1, 1))
+    dims_s, dims_b = (48, 40, 32), (128, 112, 96)
+
+    def inplace_pair(dims, seed):
+        Pi = Pencil(topo, dims, (1, 2))
+        Po = Pencil(topo, dims, (0, 2), permute=(1, 2, 0))
+        n = Pi.length_local(0)
+        buf = torch.empty(n, dtype=torch.float64, device="cuda:0")
+        g, parents = seeded_parents(dims, (1, 1), (1, 2), (0, 1, 2), (),
+                                    np.float64, seed=seed)
+        buf.copy_(_to_gpu(parents[0]))
+        src = PencilArray(Pi, 0, buf)
+        dst = PencilArray(Po, 0, buf)
+        t = Transposition(dst, src)
+        assert t.aliased
+        exp = orc.transpose_oracle(parents, dims, (1, 1), (1, 2),
+                                   (0, 1, 2), (0, 2), (1, 2, 0), ())[0]
+        return t, buf, parents[0], exp
+
+    tS, bufS, origS, expS = inplace_pair(dims_s, 111)
+    # enqueue a chain of deferred-wait executes (forward+back repeatedly,
+    # ending forward) without any sync
+    PiS = tS.src.pencil
+    back = Transposition(
+        PencilArray(PiS, 0, bufS),
+        PencilArray(tS.dest.pencil, 0, bufS))
+    for _ in range(4):
+        tS.execute(sync=False)
+        back.execute(sync=False)
+    tS.execute(sync=False)  # in flight when the big plan grows the pool
+
+    tB, bufB, origB, expB = inplace_pair(dims_b, 222)  # reserve() syncs here
+    tB.execute()
+    torch.cuda.synchronize()
+    assert np.array_equal(bufB.cpu().numpy(), expB)
+    assert np.array_equal(bufS.cpu().numpy(), expS)  # chain landed intact
+
+    # old plan re-binds to the grown pool and still bit-exact
+    bufS.copy_(_to_gpu(origS))
+    tS.execute()
+    torch.cuda.synchronize()
+    assert np.array_equal(bufS.cpu().numpy(), expS)
