@@ -514,9 +514,6 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
             constexpr int TI = 64, TJ = 128, NR = 16;
             const int64_t nti = (dn.dims[0] + TI - 1) / TI;
             const int64_t ntj = (dn.dims[ta] + TJ - 1) / TJ;
-            int64_t nbatch = 1;
-            for (int a2 = 1; a2 < dn.nd; a2++)
-                if (a2 != ta) nbatch *= dn.dims[a2];
             int jc = 1; /* same occupancy/tail-balance thresholds as below */
             if (ntj * nti * nbatch >= 32 * 4096 && ntj >= 256)
                 jc = 32;
@@ -524,7 +521,6 @@ static pa_status launch_desc(const CopyDescH &dn, int64_t esz,
                 jc = 8;
             const int64_t njc = (ntj + jc - 1) / jc;
             const int64_t nblocks = nti * njc * nbatch;
-            DescDev dd = to_dev(dn);
             dim3 grid;
             pa_status gst = grid2d(nblocks, 64 * NR, &grid);
             if (gst) return gst;
