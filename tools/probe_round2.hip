@@ -194,6 +194,55 @@ __global__ __launch_bounds__(256) void g_copy_direct_nt(
     }
 }
 
+__global__ __launch_bounds__(512) void g_copy_direct_nt512(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n) {
+        v4ug v = __builtin_nontemporal_load((const v4ug *)&src[i]);
+        __builtin_nontemporal_store(v, (v4ug *)&dst[i]);
+    }
+}
+
+/* two elements per thread (adjacent 2x16B = one 32B segment per lane) */
+__global__ __launch_bounds__(256) void g_copy_nt_x2(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = 2 * (b * blockDim.x + threadIdx.x);
+    if (i + 1 < n) {
+        v4ug v0 = __builtin_nontemporal_load((const v4ug *)&src[i]);
+        v4ug v1 = __builtin_nontemporal_load((const v4ug *)&src[i + 1]);
+        __builtin_nontemporal_store(v0, (v4ug *)&dst[i]);
+        __builtin_nontemporal_store(v1, (v4ug *)&dst[i + 1]);
+    } else if (i < n) {
+        v4ug v = __builtin_nontemporal_load((const v4ug *)&src[i]);
+        __builtin_nontemporal_store(v, (v4ug *)&dst[i]);
+    }
+}
+
+/* NT load, regular store (and vice versa): which side does the winning? */
+__global__ __launch_bounds__(256) void g_copy_ntload(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n) {
+        v4ug v = __builtin_nontemporal_load((const v4ug *)&src[i]);
+        *(v4ug *)&dst[i] = v;
+    }
+}
+
+__global__ __launch_bounds__(256) void g_copy_ntstore(
+    const uint4 *__restrict__ src, uint4 *__restrict__ dst, int64_t n)
+{
+    const int64_t b = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
+    const int64_t i = b * blockDim.x + threadIdx.x;
+    if (i < n)
+        __builtin_nontemporal_store(*(const v4ug *)&src[i], (v4ug *)&dst[i]);
+}
+
 /* ---- parameterized single-direction ceilings ---- */
 
 template <int TI, int NROWS, int JCHUNK>
@@ -569,6 +618,40 @@ int main()
              hipLaunchKernelGGL(g_copy_direct_nt, directp(n16p), dim3(256),
                                 0, 0, (const uint4 *)sp, (uint4 *)dp, n16p),
              pio);
+        if (pay == 1) { /* NT micro-variants at the 8 GiB headline payload */
+            auto direct512 = [&](int64_t n) {
+                const int64_t blocks = (n + 511) / 512;
+                const int64_t gx = blocks < 8388607 ? blocks : 8388607;
+                const int64_t gy = (blocks + gx - 1) / gx;
+                return dim3((uint32_t)gx, (uint32_t)gy);
+            };
+            auto directx2 = [&](int64_t n) {
+                const int64_t blocks = ((n + 1) / 2 + 255) / 256;
+                const int64_t gx = blocks < 16777215 ? blocks : 16777215;
+                const int64_t gy = (blocks + gx - 1) / gx;
+                return dim3((uint32_t)gx, (uint32_t)gy);
+            };
+            BEST("P1Dv nt 512thr",
+                 hipLaunchKernelGGL(g_copy_direct_nt512, direct512(n16p),
+                                    dim3(512), 0, 0, (const uint4 *)sp,
+                                    (uint4 *)dp, n16p),
+                 pio);
+            BEST("P1Dv nt x2/thread",
+                 hipLaunchKernelGGL(g_copy_nt_x2, directx2(n16p), dim3(256),
+                                    0, 0, (const uint4 *)sp, (uint4 *)dp,
+                                    n16p),
+                 pio);
+            BEST("P1Dv nt-load only",
+                 hipLaunchKernelGGL(g_copy_ntload, directp(n16p), dim3(256),
+                                    0, 0, (const uint4 *)sp, (uint4 *)dp,
+                                    n16p),
+                 pio);
+            BEST("P1Dv nt-store only",
+                 hipLaunchKernelGGL(g_copy_ntstore, directp(n16p),
+                                    dim3(256), 0, 0, (const uint4 *)sp,
+                                    (uint4 *)dp, n16p),
+                 pio);
+        }
         CHK(hipFree(sp));
         CHK(hipFree(dp));
     }
