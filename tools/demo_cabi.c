@@ -55,6 +55,22 @@ int main(void)
         fprintf(stderr, "execute failed: %s\n", pa_last_error());
         return 1;
     }
+
+    /* per-stage timing through the ABI (round-2 entry points) */
+    double st[4];
+    if (pa_plan_enable_timing(plan, 1) ||
+        pa_transpose_execute(plan, d_src, d_dst, NULL) ||
+        pa_transpose_wait(plan, NULL) || pa_plan_stage_times(plan, st)) {
+        fprintf(stderr, "stage timing failed: %s\n", pa_last_error());
+        return 1;
+    }
+    printf("demo_cabi stage_ms: pack=%.4f local=%.4f exchange=%.4f "
+           "unpack=%.4f\n", st[0], st[1], st[2], st[3]);
+    if (st[1] <= 0) { /* world-1: the fused local copy IS the step */
+        fprintf(stderr, "stage timing: local stage missing\n");
+        return 1;
+    }
+
     hipMemcpy(h_dst, d_dst, n * 8, D2H);
     hipDeviceSynchronize();
 
