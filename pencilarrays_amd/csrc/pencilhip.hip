@@ -253,7 +253,13 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile_vs(
     int64_t ntile_i, int64_t njchunk, int64_t nblocks)
 {
     static_assert(sizeof(T) == 8, "vector-store tile is 8-byte-elem only");
-    __shared__ T tile[TJ][TI + 2];
+    /* TRANSPOSED LDS layout (tile[i][j], +2 pad): the store phase reads
+     * 16 B of consecutive j per lane — one wide conflict-free LDS read.
+     * The original tile[j][i] layout measured 0.6 LDS-conflict cycles per
+     * active cycle (lane-pair row-strided reads); this layout removes them
+     * and wins the interleaved A/B by a small margin (probe vs2 p2,
+     * profiles/r2_probe_vs2*.txt). */
+    __shared__ T tile[TI][TJ + 2];
 
     const int tx = threadIdx.x;
     const int ty = threadIdx.y;
@@ -287,7 +293,7 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile_vs(
             for (int j = ty; j < nj; j += NROWS) {
                 const int64_t row = base + (int64_t)j * d.sstr[ta];
                 for (int i = tx; i < ni; i += 64)
-                    tile[j][i] = src[row + i];
+                    tile[i][j] = src[row + i];
             }
         }
         __syncthreads();
@@ -298,12 +304,12 @@ __global__ __launch_bounds__(64 * NROWS) void k_transpose_tile_vs(
                 T *row = dst + base + (int64_t)i * d.dstr[0];
                 for (int j2 = 2 * tx; j2 < njv; j2 += 128) {
                     uint4 q;
-                    ((T *)&q)[0] = tile[j2][i];
-                    ((T *)&q)[1] = tile[j2 + 1][i];
+                    ((T *)&q)[0] = tile[i][j2];
+                    ((T *)&q)[1] = tile[i][j2 + 1];
                     *(uint4 *)&row[j2] = q;
                 }
                 if ((nj & 1) && tx == 0) /* odd tail element */
-                    row[nj - 1] = tile[nj - 1][i];
+                    row[nj - 1] = tile[i][nj - 1];
             }
         }
         __syncthreads();

@@ -106,6 +106,42 @@ __global__ __launch_bounds__(64 * NROWS) void t_sweep_vs(
     }
 }
 
+/* vs with TRANSPOSED LDS layout: tile[i][j] so the store phase reads 16 B
+ * of CONSECUTIVE j per lane (one wide LDS read, no lane-pair stride
+ * conflicts); pad chosen odd-ish to spread the load phase's strided LDS
+ * writes.  PMC showed the shipped vs kernel at 0.6 LDS-conflict cycles per
+ * active cycle (store-phase row-pair reads). */
+template <int TI, int TJ, int NROWS, int JCHUNK, int PAD>
+__global__ __launch_bounds__(64 * NROWS) void t_sweep_vs2(
+    const uint64_t *__restrict__ src, uint64_t *__restrict__ dst, int64_t NI,
+    int64_t NJ, int64_t nti)
+{
+    __shared__ uint64_t tile[TI][TJ + PAD];
+    const int tx = threadIdx.x, ty = threadIdx.y;
+    const int64_t t_i = (int64_t)blockIdx.x % nti;
+    const int64_t chunk = (int64_t)blockIdx.x / nti;
+    const int64_t i0 = t_i * TI;
+    const int64_t ni = NI - i0 < TI ? NI - i0 : TI;
+    for (int jt = 0; jt < JCHUNK; jt++) {
+        const int64_t j0 = (chunk * (int64_t)JCHUNK + jt) * TJ;
+        if (j0 >= NJ) break;
+        for (int j = ty; j < TJ; j += NROWS)
+            for (int i = tx; i < ni; i += 64)
+                tile[i][j] = src[(i0 + i) + NI * (j0 + j)];
+        __syncthreads();
+        for (int i = ty; i < ni; i += NROWS) {
+            uint64_t *row = &dst[j0 + NJ * (i0 + i)];
+            for (int j2 = 2 * tx; j2 < TJ; j2 += 128) {
+                uint4 q;
+                ((uint64_t *)&q)[0] = tile[i][j2];
+                ((uint64_t *)&q)[1] = tile[i][j2 + 1];
+                *(uint4 *)&row[j2] = q;
+            }
+        }
+        __syncthreads();
+    }
+}
+
 /* scalar NONTEMPORAL stores on the production 128x64 sweep (one-line change
  * to the shipped kernel if it wins) */
 template <int TI, int TJ, int NROWS, int JCHUNK>
@@ -542,6 +578,23 @@ int main()
         RSWEEP("TRUE vs 64x128 c128", t_sweep_vs, 64, 128, 16, 128);
         RSWEEP("TRUE scalar 128x64 c32", t_sweep_c, 128, 64, 16, 32);
         RSWEEP("TRUE vsnt 64x128 c32", t_sweep_vsnt, 64, 128, 16, 32);
+#define RSWEEP5(name, TI, TJ, NR, JC, PAD)                                   \
+    {                                                                        \
+        const int64_t nti = (NIr + TI - 1) / TI;                             \
+        const int64_t ntj = (NJr + TJ - 1) / TJ;                             \
+        const int64_t njc = (ntj + JC - 1) / JC;                             \
+        BEST(name,                                                           \
+             hipLaunchKernelGGL((t_sweep_vs2<TI, TJ, NR, JC, PAD>),          \
+                                dim3((uint32_t)(nti * njc)), dim3(64, NR),   \
+                                0, 0, (const uint64_t *)s2, (uint64_t *)d2,  \
+                                NIr, NJr, nti),                              \
+             rio);                                                           \
+    }
+        RSWEEP5("TRUE vs2 p0 64x128 c32", 64, 128, 16, 32, 0);
+        RSWEEP5("TRUE vs2 p1 64x128 c32", 64, 128, 16, 32, 1);
+        RSWEEP5("TRUE vs2 p2 64x128 c32", 64, 128, 16, 32, 2);
+        RSWEEP5("TRUE vs2 p3 64x128 c32", 64, 128, 16, 32, 3);
+        RSWEEP5("TRUE vs2 p5 64x128 c32", 64, 128, 16, 32, 5);
         /* decisive interleaved A/B at THE shape: vs vs vsnt, 16 rounds */
         {
             const int ROUNDS = 16;
@@ -555,7 +608,7 @@ int main()
                                         (const uint64_t *)s2, (uint64_t *)d2,
                                         NIr, NJr, nti),
                      mV, r);
-                ONE2(hipLaunchKernelGGL((t_sweep_vsnt<64, 128, 16, 32>),
+                ONE2(hipLaunchKernelGGL((t_sweep_vs2<64, 128, 16, 32, 2>),
                                         dim3((uint32_t)(nti * njc)),
                                         dim3(64, 16), 0, 0,
                                         (const uint64_t *)s2, (uint64_t *)d2,
@@ -563,7 +616,7 @@ int main()
                      mN, r);
             }
             double *ar2[2] = {mV, mN};
-            const char *nm2[2] = {"vs  c32", "vsnt c32"};
+            const char *nm2[2] = {"vs  c32", "vs2p2c32"};
             for (int v = 0; v < 2; v++) {
                 double best = 1e30;
                 for (int r = 0; r < ROUNDS; r++)
@@ -844,6 +897,20 @@ int main()
                 for (int64_t i = 0; i < ni2; i++)
                     if (hd[j + nj2 * i] != hs[i + ni2 * j]) bad++;
             printf("correctness %-24s %s (%lld bad)\n", "vsnt 64x128 r16 c16",
+                   bad ? "FAIL" : "OK", (long long)bad);
+            CHK(hipMemset(d, 0xCC, ni2 * nj2 * 8));
+            hipLaunchKernelGGL((t_sweep_vs2<64, 128, 16, 16, 3>),
+                               dim3((uint32_t)(nti * njc)), dim3(64, 16), 0,
+                               0, (const uint64_t *)s, (uint64_t *)d, ni2,
+                               nj2, nti);
+            CHK(hipDeviceSynchronize());
+            CHK(hipGetLastError());
+            CHK(hipMemcpy(hd, d, ni2 * nj2 * 8, hipMemcpyDeviceToHost));
+            bad = 0;
+            for (int64_t j = 0; j < nj2; j++)
+                for (int64_t i = 0; i < ni2; i++)
+                    if (hd[j + nj2 * i] != hs[i + ni2 * j]) bad++;
+            printf("correctness %-24s %s (%lld bad)\n", "vs2 p3 64x128 c16",
                    bad ? "FAIL" : "OK", (long long)bad);
         }
         free(hs);
