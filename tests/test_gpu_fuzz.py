@@ -6,6 +6,7 @@ every kernel path the dispatcher can pick (vector-store tile, scalar tile,
 linear runs, 1-D copy incl. the NT gate, byte-ified odd sizes)."""
 
 import math
+import os
 
 import numpy as np
 import pytest
@@ -41,7 +42,8 @@ def test_gpu_fuzz_world1_vs_oracle():
     rng = np.random.default_rng(0xBEEFCAFE)
     DTYPES = [np.float64, np.float32, np.complex64, np.complex128,
               np.float16]
-    for trial in range(40):
+    trials = int(os.environ.get("PENCILHIP_FUZZ_TRIALS", "40"))
+    for trial in range(trials):
         nd = int(rng.integers(2, 5))
         dims = tuple(int(rng.integers(1, 40)) for _ in range(nd))
         m = int(rng.integers(1, min(nd, 3)))
@@ -94,7 +96,8 @@ def test_gpu_fuzz_inplace_world1_vs_oracle():
     (Transpositions.jl:250-264 semantics), bit-exact vs oracle."""
     rng = np.random.default_rng(0xA11A5ED)
     DTYPES = [np.float64, np.float32, np.complex64]
-    for trial in range(15):
+    trials = int(os.environ.get("PENCILHIP_FUZZ_TRIALS_INPLACE", "15"))
+    for trial in range(trials):
         nd = int(rng.integers(2, 5))
         dims = tuple(int(rng.integers(2, 32)) for _ in range(nd))
         m = int(rng.integers(1, min(nd, 3)))
