@@ -61,6 +61,11 @@ class NativePlan:
                  extra_dims: Tuple[int, ...] = (), aliased: bool = False):
         lib = self.lib = load()
         topo = Pi.topology
+        # both pa_pencils are built against ONE pa_topology handle, so the
+        # engine-side topology check can't see a host-level mismatch —
+        # validate here (assert_compatible, Transpositions.jl:184-186)
+        if tuple(Po.topology.dims) != tuple(topo.dims):
+            raise RuntimeError("pencil topologies must be the same.")
         m = topo.ndims
         self._topo = VP()
         _check(lib, lib.pa_topology_create(

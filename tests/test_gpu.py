@@ -50,6 +50,9 @@ WORLD1 = [
     ((24, 18, 12), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (3,), np.float64),
     ((24, 18, 12), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (4, 3), np.float64),
     ((1, 7, 5), (1, 2), (0, 1, 2), (0, 2), (1, 2, 0), (), np.float64),
+    # 5-D (deep descriptors on device)
+    ((5, 6, 4, 7, 3), (1, 3), (0, 1, 2, 3, 4), (0, 3), (4, 0, 2, 1, 3), (),
+     np.float64),
 ]
 
 _T_DTYPE = {

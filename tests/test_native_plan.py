@@ -103,3 +103,65 @@ def test_abi_exports_complete():
     assert len(syms) >= 20
     for s in set(syms):
         assert hasattr(lib, s), f"missing export {s}"
+
+
+def test_plan_create_error_strings_mirror_reference():
+    """assert_compatible's error messages (Transpositions.jl:182-199) are
+    kept verbatim so a host porting error-handling code sees the same
+    text."""
+    topo = Topology((2, 2))
+    Pi = Pencil(topo, (16, 21, 41), (1, 2))
+    # two-hop decomposition change
+    Po2 = Pencil(topo, (16, 21, 41), (0, 1))
+    with pytest.raises(RuntimeError,
+                       match="differ in at most one dimension"):
+        native.NativePlan(Pi, Po2, 0, 8)
+    # different global size
+    Po3 = Pencil(topo, (16, 21, 40), (0, 2))
+    with pytest.raises(RuntimeError,
+                       match="global data sizes must be the same"):
+        native.NativePlan(Pi, Po3, 0, 8)
+    # different topology
+    topo2 = Topology((4, 1))
+    Po4 = Pencil(topo2, (16, 21, 41), (0, 2))
+    with pytest.raises(RuntimeError,
+                       match="topologies must be the same"):
+        native.NativePlan(Pi, Po4, 0, 8)
+
+
+def test_pencil_create_rejects_bad_args():
+    import ctypes
+    lib = native.load()
+    I64, I32, VP = ctypes.c_int64, ctypes.c_int32, ctypes.c_void_p
+    topo = VP()
+    assert lib.pa_topology_create(2, (I64 * 2)(2, 2),
+                                  ctypes.byref(topo)) == 0
+    out = VP()
+    # repeated decomp dim (_check_selected_dimensions, Pencils.jl:393-406)
+    st = lib.pa_pencil_create(topo, 3, (I64 * 3)(8, 8, 8),
+                              (I32 * 2)(1, 1), None, ctypes.byref(out))
+    assert st != 0 and b"repeated" in lib.pa_last_error()
+    # decomp dim out of range
+    st = lib.pa_pencil_create(topo, 3, (I64 * 3)(8, 8, 8),
+                              (I32 * 2)(1, 5), None, ctypes.byref(out))
+    assert st != 0 and b"out of range" in lib.pa_last_error()
+    # invalid permutation (not a bijection)
+    st = lib.pa_pencil_create(topo, 3, (I64 * 3)(8, 8, 8),
+                              (I32 * 2)(1, 2), (I32 * 3)(0, 0, 2),
+                              ctypes.byref(out))
+    assert st != 0 and b"invalid permutation" in lib.pa_last_error()
+    # M > N
+    st = lib.pa_pencil_create(topo, 1, (I64 * 1)(8),
+                              (I32 * 2)(0, 0), None, ctypes.byref(out))
+    assert st != 0
+    lib.pa_topology_destroy(topo)
+
+
+def test_plan_create_rejects_bad_rank_and_esz():
+    topo = Topology((2, 2))
+    Pi = Pencil(topo, (16, 21, 41), (1, 2))
+    Po = Pencil(topo, (16, 21, 41), (0, 2))
+    with pytest.raises(RuntimeError, match="rank out of range"):
+        native.NativePlan(Pi, Po, 7, 8)
+    with pytest.raises(RuntimeError, match="elem_size"):
+        native.NativePlan(Pi, Po, 0, 0)

@@ -63,6 +63,11 @@ SWEEP = [
      np.float64),
     ((10, 9, 8, 7), (2, 3), (1, 2), (1, 0, 3, 2), (1, 3), (2, 3, 0, 1), (),
      np.float32),
+    # 5-D data (deep descriptors: N=5 exercises MAXND-adjacent paths)
+    ((5, 6, 4, 7, 3), (2, 2), (1, 3), (0, 1, 2, 3, 4), (0, 3),
+     (4, 0, 2, 1, 3), (), np.float64),
+    ((4, 5, 6, 3, 7), (2, 1), (2, 3), (2, 0, 4, 1, 3), (4, 3),
+     (1, 3, 0, 4, 2), (), np.complex64),
 ]
 
 
