@@ -375,3 +375,32 @@ def test_pool_growth_with_async_work_in_flight():
     tS.execute()
     torch.cuda.synchronize()
     assert np.array_equal(bufS.cpu().numpy(), expS)
+
+
+def test_gather_of_cuda_array_world1():
+    """gather on device arrays converts to host exactly (gather.jl:47: GPU
+    data converted to CPU before placement)."""
+    dims = (10, 8, 6)
+    topo = Topology((1, 1))
+    Pi = Pencil(topo, dims, (1, 2), permute=(2, 0, 1))
+    g, parents = seeded_parents(dims, (1, 1), (1, 2), (2, 0, 1), (),
+                                np.float64)
+    x = PencilArray(Pi, 0, _to_gpu(parents[0]))
+    from pencilarrays_amd import gather_sim
+    got = gather_sim([x])
+    assert got.shape == dims
+    assert np.array_equal(got, g)
+
+
+def test_set_comm_size_mismatch_rejected():
+    """pa_plan_set_comm validates the communicator against the subgroup
+    (nranks == P, rank == my coordinate along R)."""
+    uid = native._nccl_uid()
+    comm = native.NativeComm.create(uid, 1, 0)
+    topo = Topology((2, 1))
+    Pi = Pencil(topo, (16, 12, 8), (1, 2))
+    Po = Pencil(topo, (16, 12, 8), (0, 2))
+    nat = native.NativePlan(Pi, Po, 0, 8)
+    assert nat.nproc_sub == 2
+    with pytest.raises(RuntimeError, match="comm size 1 != subgroup size 2"):
+        nat.set_comm(comm)
