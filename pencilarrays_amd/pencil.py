@@ -106,6 +106,22 @@ class Pencil:
                            else check_perm(permute))
         self.ndims = n
 
+    @classmethod
+    def from_nprocs(cls, size_global: Sequence[int], nprocs: int,
+                    decomp_dims: Optional[Sequence[int]] = None,
+                    permute: Optional[Sequence[int]] = None) -> "Pencil":
+        """``Pencil(size_global, [decomp_dims,] comm)`` (Pencils.jl:106-116):
+        the convenience constructor that creates the process topology
+        implicitly — `MPI.Dims_create` over the decomposed dimensions
+        (default: dimensions 1..N-1 0-based, i.e. the reference's 2:N)."""
+        n = len(size_global)
+        if decomp_dims is None:
+            decomp_dims = tuple(range(1, n))
+        m = len(decomp_dims)
+        from .topology import dims_create
+        topo = Topology(dims_create(nprocs, m))
+        return cls(topo, size_global, decomp_dims, permute=permute)
+
     # ---- axes ----------------------------------------------------------
 
     def axes_for_coords(self, coords: Sequence[int]) -> Region:

@@ -111,3 +111,19 @@ def test_dims_create():
     assert dims_create(7, 2) == (7, 1)
     for n, m in [(8, 2), (12, 3), (5, 2)]:
         assert math.prod(dims_create(n, m)) == n
+
+
+def test_from_nprocs_convenience():
+    """Pencil(dims, comm) convenience form (Pencils.jl:106-116): implicit
+    topology via Dims_create over the default decomp dims 2:N (0-based
+    1..N-1), mirroring the doctest `Pencil((4, 8, 12), MPI.COMM_WORLD)` ->
+    decomposed dimensions (2, 3)."""
+    from pencilarrays_amd import Pencil
+    p = Pencil.from_nprocs((4, 8, 12), 4)
+    assert p.decomp_dims == (1, 2)          # Julia (2, 3)
+    assert tuple(p.topology.dims) == (2, 2)  # Dims_create(4, 2)
+    assert p.topology.nranks == 4
+    # explicit single decomposed dim: Pencil((4, 8, 12), (1,), comm)
+    p2 = Pencil.from_nprocs((4, 8, 12), 3, decomp_dims=(0,))
+    assert p2.decomp_dims == (0,)
+    assert tuple(p2.topology.dims) == (3,)
