@@ -33,6 +33,9 @@ sys.path.insert(0, REPO)
 # processes fails with hipIpcGetMemHandle errors (normally exported by the
 # environment — kept here as insurance for the multi-process run)
 os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+# surface RCCL warnings/errors in stderr for the unattended N>1 run
+# (default VERSION level prints only the banner)
+os.environ.setdefault("NCCL_DEBUG", "WARN")
 
 import torch  # noqa: E402
 
