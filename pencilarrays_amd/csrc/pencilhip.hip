@@ -33,7 +33,10 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <sched.h>
+
 #include <algorithm>
+#include <chrono>
 #include <cstdarg>
 #include <cstdio>
 #include <cstdlib>
@@ -1057,10 +1060,49 @@ pa_status pa_comm_create(const char *id, int nranks, int rank, pa_comm **out)
     c->rank = rank;
     ncclUniqueId uid;
     memcpy(&uid, id, sizeof uid);
-    ncclResult_t r = ncclCommInitRank(&c->comm, nranks, uid, rank);
-    if (r != ncclSuccess) {
+    /* Non-blocking init with a bounded wait: a blocking ncclCommInitRank
+     * hangs forever if any subgroup rank is missing or disagrees — in an
+     * unattended multi-process run that is an undiagnosable timeout.
+     * Timeout via PENCILHIP_COMM_TIMEOUT_S (default 180 s). */
+    ncclConfig_t cfg = NCCL_CONFIG_INITIALIZER;
+    cfg.blocking = 0;
+    ncclResult_t r = ncclCommInitRankConfig(&c->comm, nranks, uid, rank, &cfg);
+    if (r != ncclSuccess && r != ncclInProgress) {
         delete c;
         return fail("ncclCommInitRank: %s", ncclGetErrorString(r));
+    }
+    double timeout_s = 180.0;
+    if (const char *te = getenv("PENCILHIP_COMM_TIMEOUT_S")) {
+        double v = atof(te);
+        if (v > 0) timeout_s = v;
+    }
+    const auto t0 = std::chrono::steady_clock::now();
+    ncclResult_t st = ncclInProgress;
+    while (true) {
+        ncclResult_t qr = ncclCommGetAsyncError(c->comm, &st);
+        if (qr != ncclSuccess) {
+            (void)ncclCommAbort(c->comm);
+            delete c;
+            return fail("ncclCommGetAsyncError: %s", ncclGetErrorString(qr));
+        }
+        if (st == ncclSuccess) break;
+        if (st != ncclInProgress) {
+            (void)ncclCommAbort(c->comm);
+            delete c;
+            return fail("ncclCommInitRank (async): %s",
+                        ncclGetErrorString(st));
+        }
+        const double waited = std::chrono::duration<double>(
+            std::chrono::steady_clock::now() - t0).count();
+        if (waited > timeout_s) {
+            (void)ncclCommAbort(c->comm);
+            delete c;
+            return fail("ncclCommInitRank timed out after %.0f s (subgroup "
+                        "nranks=%d rank=%d): a peer is missing or "
+                        "disagrees on the unique id", timeout_s, nranks,
+                        rank);
+        }
+        sched_yield();
     }
     *out = c;
     return 0;
